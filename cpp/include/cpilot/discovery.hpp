@@ -1,0 +1,117 @@
+// Consul service-discovery backend: registration, TTL heartbeats,
+// deregistration, and polled health queries with compare-and-swap change
+// detection. Blocking HTTP runs on a small worker pool; results are posted
+// back onto the reactor so all state stays single-threaded.
+// Parity: /root/reference/discovery/{discovery,consul,config,service}.go.
+#pragma once
+
+#include <condition_variable>
+#include <deque>
+#include <functional>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "cpilot/json.hpp"
+#include "cpilot/loop.hpp"
+#include "cpilot/metrics.hpp"
+
+namespace cpilot {
+
+struct ServiceEntry {
+  std::string id;
+  std::string address;
+  int port = 0;
+};
+
+class ConsulBackend {
+ public:
+  // Build from the top-level "consul" config value: a URI string or a map
+  // with {address, scheme, token, tls:{...}}. CONSUL_HTTP_TOKEN overrides
+  // the token (discovery/consul.go:48-50). Returns nullptr + err on
+  // "no discovery backend defined".
+  static std::unique_ptr<ConsulBackend> create(const Json* raw,
+                                               std::string* err);
+
+  ~ConsulBackend();
+
+  void start(Loop& loop);
+  void stop();
+
+  using DoneCb = std::function<void(bool ok, const std::string& err)>;
+  using HealthCb =
+      std::function<void(bool ok, std::vector<ServiceEntry> entries)>;
+
+  // byte-compatible payload with the reference's registration
+  // (discovery/service.go:93-110)
+  void serviceRegister(const std::string& id, const std::string& name,
+                       const std::vector<std::string>& tags, int port,
+                       const std::string& address, bool enableTagOverride,
+                       int ttlSeconds, const std::string& status,
+                       const std::string& deregisterAfter, DoneCb cb);
+  void updateTTL(const std::string& checkID, const std::string& output,
+                 const std::string& status, DoneCb cb);
+  void serviceDeregister(const std::string& id, DoneCb cb);
+  // GET /v1/health/service/<name>?passing=1 (+tag,+dc)
+  void healthService(const std::string& name, const std::string& tag,
+                     const std::string& dc, HealthCb cb);
+
+  // change detection against the cached set (loop thread only)
+  // (discovery/consul.go:102-125)
+  bool compareAndSwap(const std::string& service,
+                      std::vector<ServiceEntry> entries);
+
+  const std::string& address() const { return address_; }
+  const std::string& scheme() const { return scheme_; }
+
+  std::shared_ptr<prom::Family> watchGauge() { return watchGauge_; }
+
+ private:
+  ConsulBackend() = default;
+  void workerMain();
+  void enqueue(std::function<void()> task);
+
+  std::string address_;  // host:port
+  std::string scheme_ = "http";
+  std::string token_;
+
+  Loop* loop_ = nullptr;
+  std::vector<std::thread> workers_;
+  std::deque<std::function<void()>> tasks_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  bool stopping_ = false;
+
+  std::map<std::string, std::vector<ServiceEntry>> watched_;
+  std::shared_ptr<prom::Family> watchGauge_;
+};
+
+// Per-job service registration state (discovery/service.go:12-110).
+struct ServiceDefinition {
+  std::string id;
+  std::string name;
+  int port = 0;
+  int ttl = 0;
+  std::vector<std::string> tags;
+  std::string initialStatus;
+  std::string ipAddress;
+  bool enableTagOverride = false;
+  std::string deregisterCriticalServiceAfter;
+  ConsulBackend* consul = nullptr;
+
+  bool wasRegistered = false;
+  bool registerInFlight = false;
+
+  void sendHeartbeat();
+  void registerWithInitialStatus();
+  void deregister();
+  void markForMaintenance();
+
+ private:
+  void registerService(const std::string& status);
+};
+
+}  // namespace cpilot

@@ -1,0 +1,123 @@
+// Event model and bus.
+//
+// Event{code, source} with the reference's 17-code enum and config-string
+// mapping (events/events.go:21-86). The Bus is a strict-FIFO dispatcher:
+// publish() appends to a queue that the reactor drains, delivering each
+// event to every live subscriber in subscription order — the
+// single-threaded equivalent of the reference's fan-out into buffered
+// channels (events/bus.go:125-140), with the same 10-slot debug ring and
+// reload flag, plus a dispatch-latency histogram for the perf baseline.
+#pragma once
+
+#include <deque>
+#include <functional>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "cpilot/loop.hpp"
+#include "cpilot/metrics.hpp"
+
+namespace cpilot {
+
+enum class EventCode {
+  None = 0,
+  ExitSuccess,
+  ExitFailed,
+  Stopping,
+  Stopped,
+  StatusHealthy,
+  StatusUnhealthy,
+  StatusChanged,
+  TimerExpired,
+  EnterMaintenance,
+  ExitMaintenance,
+  Error,
+  Quit,
+  Metric,
+  Startup,
+  Shutdown,
+  Signal,
+};
+
+const char* eventCodeString(EventCode code);
+// Parse a config event-name ("exitSuccess", "healthy", "SIGHUP", ...);
+// returns false if invalid. (events/events.go:52-86)
+bool eventCodeFromString(const std::string& name, EventCode* out);
+
+struct Event {
+  EventCode code = EventCode::None;
+  std::string source;
+
+  bool operator==(const Event& o) const {
+    return code == o.code && source == o.source;
+  }
+  bool operator!=(const Event& o) const { return !(*this == o); }
+  std::string str() const {
+    return std::string("{") + eventCodeString(code) + " " + source + "}";
+  }
+};
+
+// global sentinel events (events/events.go:42-49)
+extern const Event GlobalStartup;
+extern const Event GlobalShutdown;
+extern const Event NonEvent;
+extern const Event GlobalEnterMaintenance;
+extern const Event GlobalExitMaintenance;
+extern const Event QuitByTest;
+
+class Subscriber {
+ public:
+  virtual ~Subscriber() = default;
+  virtual void onEvent(const Event& event) = 0;
+};
+
+class Bus {
+ public:
+  explicit Bus(Loop& loop);
+
+  void subscribe(Subscriber* s);
+  void unsubscribe(Subscriber* s);
+
+  void publish(Event event);
+  void publishSignal(const std::string& sig) {
+    publish(Event{EventCode::Signal, sig});
+  }
+  void shutdown() { publish(GlobalShutdown); }
+  void setReloadFlag() { reload_ = true; }
+  bool reloadFlag() const { return reload_; }
+
+  // drain the debug ring (tests; events/bus.go:33-54)
+  std::vector<Event> debugEvents();
+
+  // bench instrumentation
+  uint64_t publishedCount() const { return published_; }
+  uint64_t deliveredCount() const { return delivered_; }
+  // dispatch latencies in seconds, via prometheus histogram family
+  std::shared_ptr<prom::Family> dispatchHist() const { return dispatchHist_; }
+  // raw latency samples ring (for exact p99 in the stats report)
+  const std::vector<double>& latencyWindow() const { return latencyWindow_; }
+
+  Loop& loop() { return loop_; }
+
+ private:
+  void drain();
+
+  Loop& loop_;
+  std::vector<Subscriber*> subscribers_;
+  std::deque<std::pair<Event, TimePoint>> queue_;
+  bool drainScheduled_ = false;
+  bool reload_ = false;
+
+  // debug ring of 10
+  std::vector<Event> ring_;
+  int head_ = -1, tail_ = 0;
+
+  uint64_t published_ = 0, delivered_ = 0;
+  std::shared_ptr<prom::Family> eventsCounter_;
+  std::shared_ptr<prom::Family> dispatchHist_;
+  std::vector<double> latencyWindow_;
+  size_t latencyCap_ = 262144;
+};
+
+}  // namespace cpilot

@@ -1,0 +1,107 @@
+// Single-threaded epoll reactor: fd watchers, a timer heap backed by one
+// timerfd, deferred callbacks, cross-thread task posting, and a child
+// process table reaped on SIGCHLD.
+//
+// This replaces the reference's goroutine-per-component model
+// (GOMAXPROCS=1, main.go:19): one reactor delivers the same observable
+// semantics (every component sees every event in publish order) without
+// locks on the hot path.
+#pragma once
+
+#include <sys/types.h>
+
+#include <chrono>
+#include <cstdint>
+#include <deque>
+#include <functional>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <queue>
+#include <vector>
+
+namespace cpilot {
+
+using Clock = std::chrono::steady_clock;
+using TimePoint = Clock::time_point;
+using Ns = std::chrono::nanoseconds;
+
+class Loop {
+ public:
+  using FdCallback = std::function<void(uint32_t epollEvents)>;
+  using TimerCallback = std::function<void()>;
+  using ChildCallback = std::function<void(int waitStatus)>;
+
+  Loop();
+  ~Loop();
+
+  Loop(const Loop&) = delete;
+  Loop& operator=(const Loop&) = delete;
+
+  // --- fds ---
+  void watchFd(int fd, uint32_t events, FdCallback cb);
+  void modifyFd(int fd, uint32_t events);
+  void unwatchFd(int fd);
+
+  // --- timers (ids are never 0) ---
+  uint64_t addTimeout(Ns delay, TimerCallback cb);
+  uint64_t addInterval(Ns interval, TimerCallback cb);
+  void cancelTimer(uint64_t id);
+
+  // --- deferred work (runs on the next loop iteration, FIFO) ---
+  void defer(std::function<void()> fn);
+
+  // --- cross-thread: safe to call from worker threads ---
+  void post(std::function<void()> fn);
+
+  // --- children ---
+  void watchChild(pid_t pid, ChildCallback cb);
+  // call on SIGCHLD: waitpid(-1, WNOHANG) loop dispatching to callbacks
+  void reapChildren();
+
+  void run();   // until stop()
+  void stop();
+
+  bool stopped() const { return stopped_; }
+
+ private:
+  struct Timer {
+    uint64_t id;
+    TimePoint deadline;
+    Ns interval;  // zero for one-shot
+    TimerCallback cb;
+    bool canceled = false;
+  };
+  struct TimerCmp {
+    bool operator()(const std::shared_ptr<Timer>& a,
+                    const std::shared_ptr<Timer>& b) const {
+      return a->deadline > b->deadline;
+    }
+  };
+
+  void armTimerFd();
+  void fireDueTimers();
+  void drainDeferred();
+  void drainPosted();
+
+  int epfd_ = -1;
+  int timerfd_ = -1;
+  int wakeupFds_[2] = {-1, -1};  // pipe for post()
+  bool stopped_ = false;
+
+  std::map<int, FdCallback> fdCallbacks_;
+  std::priority_queue<std::shared_ptr<Timer>,
+                      std::vector<std::shared_ptr<Timer>>, TimerCmp>
+      timers_;
+  std::map<uint64_t, std::shared_ptr<Timer>> timersById_;
+  uint64_t nextTimerId_ = 1;
+
+  std::deque<std::function<void()>> deferred_;
+
+  std::mutex postedMu_;
+  std::deque<std::function<void()>> posted_;
+
+  std::map<pid_t, ChildCallback> children_;
+};
+
+}  // namespace cpilot

@@ -1,0 +1,279 @@
+#include "cpilot/command.hpp"
+
+#include <fcntl.h>
+#include <signal.h>
+#include <sys/epoll.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <cctype>
+#include <cstring>
+
+#include "cpilot/log.hpp"
+
+namespace cpilot {
+
+bool parseArgs(const Json& raw, std::string* execPath,
+               std::vector<std::string>* args, std::string* err) {
+  std::vector<std::string> all;
+  if (raw.isString()) {
+    // TrimSpace then split on single spaces (commands/args.go:14-18)
+    std::string s = raw.str();
+    size_t a = 0, b = s.size();
+    while (a < b && isspace((unsigned char)s[a])) a++;
+    while (b > a && isspace((unsigned char)s[b - 1])) b--;
+    s = s.substr(a, b - a);
+    if (!s.empty()) {
+      std::string cur;
+      for (char c : s) {
+        if (c == ' ') {
+          all.push_back(cur);
+          cur.clear();
+        } else {
+          cur += c;
+        }
+      }
+      all.push_back(cur);
+    }
+  } else if (raw.isArray()) {
+    for (auto& e : raw.array()) {
+      if (e.isString()) all.push_back(e.str());
+      else all.push_back(e.dump());
+    }
+  } else if (!raw.isNull()) {
+    *err = "unable to parse exec arguments";
+    return false;
+  }
+  if (all.empty()) {
+    *err = "received zero-length argument";
+    return false;
+  }
+  *execPath = all[0];
+  args->assign(all.begin() + 1, all.end());
+  return true;
+}
+
+CommandPtr newCommand(const Json& rawExec, Duration timeout, bool raw,
+                      const std::string& logField, std::string* err) {
+  std::string execPath;
+  std::vector<std::string> args;
+  if (!parseArgs(rawExec, &execPath, &args, err)) return nullptr;
+  return std::make_shared<Command>(execPath, args, timeout, raw, logField);
+}
+
+Command::Command(std::string execPath, std::vector<std::string> args,
+                 Duration timeout, bool raw, std::string logField)
+    : name_(execPath),
+      exec_(std::move(execPath)),
+      args_(std::move(args)),
+      timeout_(timeout),
+      raw_(raw),
+      logField_(std::move(logField)) {}
+
+std::string Command::envName() const {
+  if (name_.empty()) return name_;
+  // base name
+  std::string name = name_;
+  size_t slash = name.find_last_of('/');
+  if (slash != std::string::npos) name = name.substr(slash + 1);
+  // strip one extension
+  size_t dot = name.find_last_of('.');
+  if (dot != std::string::npos && dot > 0)
+    name = name.substr(0, dot) + name.substr(name.size());
+  // non-alphanumerics -> underscore, compact doubles
+  std::string out;
+  for (char c : name) {
+    if (isalnum((unsigned char)c)) {
+      out += (char)toupper((unsigned char)c);
+    } else if (out.empty() || out.back() != '_') {
+      out += '_';
+    }
+  }
+  return out;
+}
+
+void Command::run(Loop& loop, std::shared_ptr<Bus> bus) {
+  if (running_) {
+    // the reference queues concurrent runs on a mutex
+    // (commands/commands.go:92-93); we bound that queue at one
+    LOG_DEBUG("%s already running, queueing run", name_.c_str());
+    pendingRun_ = true;
+    return;
+  }
+  spawn(loop, std::move(bus));
+}
+
+void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
+  loop_ = &loop;
+  LOG_DEBUG("%s.Run start", name_.c_str());
+
+  int pipefds[2] = {-1, -1};
+  if (!raw_) {
+    if (pipe2(pipefds, O_CLOEXEC) != 0) {
+      LOG_ERROR("unable to create pipe for %s: %s", name_.c_str(),
+                strerror(errno));
+      bus->publish(Event{EventCode::ExitFailed, name_});
+      bus->publish(Event{EventCode::Error, strerror(errno)});
+      return;
+    }
+  }
+
+  pid_t pid = fork();
+  if (pid < 0) {
+    LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(errno));
+    if (pipefds[0] >= 0) {
+      close(pipefds[0]);
+      close(pipefds[1]);
+    }
+    bus->publish(Event{EventCode::ExitFailed, name_});
+    bus->publish(Event{EventCode::Error, strerror(errno)});
+    return;
+  }
+  if (pid == 0) {
+    // child
+    setpgid(0, 0);
+    if (!raw_) {
+      dup2(pipefds[1], 1);
+      dup2(pipefds[1], 2);
+    }
+    // unblock all signals (the daemon blocks them for signalfd)
+    sigset_t empty;
+    sigemptyset(&empty);
+    sigprocmask(SIG_SETMASK, &empty, nullptr);
+    std::vector<char*> argv;
+    argv.push_back(const_cast<char*>(exec_.c_str()));
+    for (auto& a : args_) argv.push_back(const_cast<char*>(a.c_str()));
+    argv.push_back(nullptr);
+    execvp(exec_.c_str(), argv.data());
+    fprintf(stderr, "unable to start %s: %s\n", exec_.c_str(),
+            strerror(errno));
+    _exit(127);
+  }
+
+  // parent
+  setpgid(pid, pid);  // avoid race with child's setpgid
+  pid_ = pid;
+  running_ = true;
+
+  std::string pidEnv = "CONTAINERPILOT_" + envName() + "_PID";
+  setenv(pidEnv.c_str(), std::to_string(pid).c_str(), 1);
+
+  if (!raw_) {
+    close(pipefds[1]);
+    logFd_ = pipefds[0];
+    fcntl(logFd_, F_SETFL, O_NONBLOCK);
+    auto self = shared_from_this();
+    loop.watchFd(logFd_, EPOLLIN | EPOLLHUP, [this, self](uint32_t) {
+      char buf[4096];
+      while (true) {
+        ssize_t n = read(logFd_, buf, sizeof(buf));
+        if (n > 0) {
+          logBuf_.append(buf, n);
+          size_t pos;
+          while ((pos = logBuf_.find('\n')) != std::string::npos) {
+            logging::logFields(logging::Level::Info, logField_, pid_,
+                               logBuf_.substr(0, pos));
+            logBuf_.erase(0, pos + 1);
+          }
+        } else if (n == 0 || (n < 0 && errno != EAGAIN)) {
+          if (!logBuf_.empty()) {
+            logging::logFields(logging::Level::Info, logField_, pid_, logBuf_);
+            logBuf_.clear();
+          }
+          loop_->unwatchFd(logFd_);
+          close(logFd_);
+          logFd_ = -1;
+          return;
+        } else {
+          return;  // EAGAIN
+        }
+      }
+    });
+  }
+
+  if (timeout_ > Duration(0)) {
+    auto self = shared_from_this();
+    timeoutTimer_ = loop.addTimeout(timeout_, [this, self] {
+      LOG_WARN("%s timeout after %llds", name_.c_str(),
+               (long long)std::chrono::duration_cast<std::chrono::seconds>(
+                   timeout_)
+                   .count());
+      timeoutTimer_ = 0;
+      kill();
+    });
+  }
+
+  auto self = shared_from_this();
+  loop.watchChild(pid, [this, self, bus](int status) {
+    onExit(*loop_, bus, status);
+  });
+}
+
+void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
+  LOG_DEBUG("%s.Run end", name_.c_str());
+  if (timeoutTimer_) {
+    loop.cancelTimer(timeoutTimer_);
+    timeoutTimer_ = 0;
+  }
+  if (logFd_ >= 0) {
+    // drain any remaining output synchronously
+    char buf[4096];
+    ssize_t n;
+    while ((n = read(logFd_, buf, sizeof(buf))) > 0) logBuf_.append(buf, n);
+    size_t pos;
+    while ((pos = logBuf_.find('\n')) != std::string::npos) {
+      logging::logFields(logging::Level::Info, logField_, pid_,
+                         logBuf_.substr(0, pos));
+      logBuf_.erase(0, pos + 1);
+    }
+    if (!logBuf_.empty()) {
+      logging::logFields(logging::Level::Info, logField_, pid_, logBuf_);
+      logBuf_.clear();
+    }
+    loop.unwatchFd(logFd_);
+    close(logFd_);
+    logFd_ = -1;
+  }
+
+  std::string pidEnv = "CONTAINERPILOT_" + envName() + "_PID";
+  unsetenv(pidEnv.c_str());
+
+  running_ = false;
+  pid_ = -1;
+
+  bool success = WIFEXITED(status) && WEXITSTATUS(status) == 0;
+  if (success) {
+    LOG_DEBUG("%s exited without error", name_.c_str());
+    bus->publish(Event{EventCode::ExitSuccess, name_});
+  } else {
+    int code = WIFEXITED(status) ? WEXITSTATUS(status)
+                                 : 128 + (WIFSIGNALED(status) ? WTERMSIG(status) : 0);
+    LOG_ERROR("%s exited with error: exit status %d", name_.c_str(), code);
+    bus->publish(Event{EventCode::ExitFailed, name_});
+    bus->publish(Event{EventCode::Error,
+                      name_ + ": exit status " + std::to_string(code)});
+  }
+
+  if (pendingRun_) {
+    pendingRun_ = false;
+    spawn(loop, bus);
+  }
+}
+
+void Command::term() {
+  LOG_DEBUG("%s.term", name_.c_str());
+  if (pid_ > 0) {
+    LOG_DEBUG("terminating command '%s' at pid: %d", name_.c_str(), pid_);
+    ::kill(-pid_, SIGTERM);
+  }
+}
+
+void Command::kill() {
+  LOG_DEBUG("%s.kill", name_.c_str());
+  if (pid_ > 0) {
+    LOG_DEBUG("killing command '%s' at pid: %d", name_.c_str(), pid_);
+    ::kill(-pid_, SIGKILL);
+  }
+}
+
+}  // namespace cpilot

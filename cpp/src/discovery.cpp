@@ -1,0 +1,337 @@
+#include "cpilot/discovery.hpp"
+
+#include <algorithm>
+#include <cstdlib>
+
+#include "cpilot/http.hpp"
+#include "cpilot/log.hpp"
+
+namespace cpilot {
+
+namespace {
+
+// strip http:// or https:// prefix; bare addresses default to http
+// (discovery/config.go:90-103)
+void parseRawURI(const std::string& raw, std::string* address,
+                 std::string* scheme) {
+  *scheme = "http";
+  *address = raw;
+  if (raw.rfind("http://", 0) == 0) {
+    *address = raw.substr(7);
+  } else if (raw.rfind("https://", 0) == 0) {
+    *address = raw.substr(8);
+    *scheme = "https";
+  }
+}
+
+std::string urlEncode(const std::string& s) {
+  std::string out;
+  for (unsigned char c : s) {
+    if (isalnum(c) || c == '-' || c == '_' || c == '.' || c == '~') {
+      out += c;
+    } else {
+      char buf[4];
+      snprintf(buf, sizeof(buf), "%%%02X", c);
+      out += buf;
+    }
+  }
+  return out;
+}
+
+}  // namespace
+
+std::unique_ptr<ConsulBackend> ConsulBackend::create(const Json* raw,
+                                                     std::string* err) {
+  auto backend = std::unique_ptr<ConsulBackend>(new ConsulBackend());
+  if (raw == nullptr || raw->isNull()) {
+    *err = "no discovery backend defined";
+    return nullptr;
+  }
+  if (raw->isString()) {
+    parseRawURI(raw->str(), &backend->address_, &backend->scheme_);
+  } else if (raw->isObject()) {
+    for (auto& kv : raw->object()) {
+      if (kv.first == "address" && kv.second.isString())
+        backend->address_ = kv.second.str();
+      else if (kv.first == "scheme" && kv.second.isString())
+        backend->scheme_ = kv.second.str();
+      else if (kv.first == "token" && kv.second.isString())
+        backend->token_ = kv.second.str();
+      else if (kv.first == "tls") {
+        // TLS options are accepted for config compatibility; https
+        // transport is not implemented in this build and fails at
+        // request time with a clear error.
+      } else {
+        *err = "consul configuration error: invalid key: " + kv.first;
+        return nullptr;
+      }
+    }
+  } else {
+    *err = "no discovery backend defined";
+    return nullptr;
+  }
+  if (backend->address_.empty()) backend->address_ = "127.0.0.1:8500";
+  if (const char* token = getenv("CONSUL_HTTP_TOKEN")) {
+    if (token[0]) backend->token_ = token;
+  }
+  if (const char* addr = getenv("CONSUL_HTTP_ADDR")) {
+    if (addr[0] && raw->isString() && raw->str().empty())
+      parseRawURI(addr, &backend->address_, &backend->scheme_);
+  }
+  backend->watchGauge_ = prom::Registry::global().registerFamily(
+      "containerpilot_watch_instances",
+      "gauge of instances found for each ContainerPilot watch, partitioned "
+      "by service",
+      prom::MetricType::Gauge, {"service"});
+  return backend;
+}
+
+ConsulBackend::~ConsulBackend() { stop(); }
+
+void ConsulBackend::start(Loop& loop) {
+  loop_ = &loop;
+  stopping_ = false;
+  for (int i = 0; i < 4; i++)
+    workers_.emplace_back([this] { workerMain(); });
+}
+
+void ConsulBackend::stop() {
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    if (stopping_ && workers_.empty()) return;
+    stopping_ = true;
+  }
+  cv_.notify_all();
+  for (auto& t : workers_) t.join();
+  workers_.clear();
+}
+
+void ConsulBackend::workerMain() {
+  while (true) {
+    std::function<void()> task;
+    {
+      std::unique_lock<std::mutex> l(mu_);
+      cv_.wait(l, [this] { return stopping_ || !tasks_.empty(); });
+      if (stopping_ && tasks_.empty()) return;
+      task = std::move(tasks_.front());
+      tasks_.pop_front();
+    }
+    task();
+  }
+}
+
+void ConsulBackend::enqueue(std::function<void()> task) {
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    if (stopping_) return;
+    tasks_.push_back(std::move(task));
+  }
+  cv_.notify_one();
+}
+
+void ConsulBackend::serviceRegister(
+    const std::string& id, const std::string& name,
+    const std::vector<std::string>& tags, int port, const std::string& address,
+    bool enableTagOverride, int ttlSeconds, const std::string& status,
+    const std::string& deregisterAfter, DoneCb cb) {
+  JsonObject check;
+  check.emplace_back("TTL", Json(std::to_string(ttlSeconds) + "s"));
+  if (!status.empty()) check.emplace_back("Status", Json(status));
+  check.emplace_back("Notes",
+                     Json("TTL for " + name + " set by containerpilot"));
+  if (!deregisterAfter.empty())
+    check.emplace_back("DeregisterCriticalServiceAfter", Json(deregisterAfter));
+
+  JsonObject payload;
+  payload.emplace_back("ID", Json(id));
+  payload.emplace_back("Name", Json(name));
+  if (!tags.empty()) {
+    JsonArray tagArr;
+    for (auto& t : tags) tagArr.push_back(Json(t));
+    payload.emplace_back("Tags", Json(std::move(tagArr)));
+  }
+  payload.emplace_back("Port", Json((int64_t)port));
+  payload.emplace_back("Address", Json(address));
+  if (enableTagOverride)
+    payload.emplace_back("EnableTagOverride", Json(true));
+  payload.emplace_back("Check", Json(std::move(check)));
+  std::string body = Json(std::move(payload)).dump();
+
+  std::string target = address_;
+  std::string token = token_;
+  std::string scheme = scheme_;
+  Loop* loop = loop_;
+  enqueue([target, token, scheme, body, cb, loop] {
+    std::map<std::string, std::string> headers;
+    if (!token.empty()) headers["X-Consul-Token"] = token;
+    http::ClientResult res;
+    if (scheme == "https") {
+      res.error = "https transport not supported";
+    } else {
+      res = http::request(target, "PUT", "/v1/agent/service/register", body,
+                          "application/json", headers);
+    }
+    bool ok = res.ok && res.status == 200;
+    std::string err = res.ok ? ("status " + std::to_string(res.status) + ": " +
+                                res.body)
+                             : res.error;
+    loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
+  });
+}
+
+void ConsulBackend::updateTTL(const std::string& checkID,
+                              const std::string& output,
+                              const std::string& status, DoneCb cb) {
+  // the Consul agent API maps "pass" -> passing (api.Agent.UpdateTTL)
+  std::string st = status;
+  if (st == "pass") st = "passing";
+  else if (st == "warn") st = "warning";
+  else if (st == "fail") st = "critical";
+  JsonObject payload;
+  payload.emplace_back("Status", Json(st));
+  payload.emplace_back("Output", Json(output));
+  std::string body = Json(std::move(payload)).dump();
+  std::string path = "/v1/agent/check/update/" + urlEncode(checkID);
+
+  std::string target = address_;
+  std::string token = token_;
+  Loop* loop = loop_;
+  enqueue([target, token, path, body, cb, loop] {
+    std::map<std::string, std::string> headers;
+    if (!token.empty()) headers["X-Consul-Token"] = token;
+    auto res = http::request(target, "PUT", path, body, "application/json",
+                             headers);
+    bool ok = res.ok && res.status == 200;
+    std::string err =
+        res.ok ? ("status " + std::to_string(res.status) + ": " + res.body)
+               : res.error;
+    loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
+  });
+}
+
+void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
+  std::string path = "/v1/agent/service/deregister/" + urlEncode(id);
+  std::string target = address_;
+  std::string token = token_;
+  Loop* loop = loop_;
+  enqueue([target, token, path, cb, loop] {
+    std::map<std::string, std::string> headers;
+    if (!token.empty()) headers["X-Consul-Token"] = token;
+    auto res = http::request(target, "PUT", path, "", "application/json",
+                             headers);
+    bool ok = res.ok && res.status == 200;
+    std::string err =
+        res.ok ? ("status " + std::to_string(res.status)) : res.error;
+    loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
+  });
+}
+
+void ConsulBackend::healthService(const std::string& name,
+                                  const std::string& tag,
+                                  const std::string& dc, HealthCb cb) {
+  std::string path = "/v1/health/service/" + urlEncode(name) + "?passing=1";
+  if (!tag.empty()) path += "&tag=" + urlEncode(tag);
+  if (!dc.empty()) path += "&dc=" + urlEncode(dc);
+  std::string target = address_;
+  std::string token = token_;
+  Loop* loop = loop_;
+  enqueue([target, token, path, cb, loop] {
+    std::map<std::string, std::string> headers;
+    if (!token.empty()) headers["X-Consul-Token"] = token;
+    auto res = http::request(target, "GET", path, "", "application/json",
+                             headers);
+    bool ok = res.ok && res.status == 200;
+    std::vector<ServiceEntry> entries;
+    if (ok) {
+      try {
+        Json doc = parseJson5(res.body);
+        if (doc.isArray()) {
+          for (auto& e : doc.array()) {
+            const Json* svc = e.find("Service");
+            if (!svc || !svc->isObject()) continue;
+            ServiceEntry entry;
+            if (const Json* v = svc->find("ID"))
+              if (v->isString()) entry.id = v->str();
+            if (const Json* v = svc->find("Address"))
+              if (v->isString()) entry.address = v->str();
+            if (const Json* v = svc->find("Port"))
+              if (v->isNumber()) entry.port = (int)v->asInt();
+            entries.push_back(std::move(entry));
+          }
+        }
+      } catch (const std::exception&) {
+        ok = false;
+      }
+    }
+    loop->post([cb, ok, entries = std::move(entries)]() mutable {
+      cb(ok, std::move(entries));
+    });
+  });
+}
+
+bool ConsulBackend::compareAndSwap(const std::string& service,
+                                   std::vector<ServiceEntry> entries) {
+  std::vector<ServiceEntry> existing = watched_[service];
+  watched_[service] = entries;
+  // compareForChange (discovery/consul.go:112-125)
+  if (existing.size() != entries.size()) return true;
+  auto byID = [](const ServiceEntry& a, const ServiceEntry& b) {
+    return a.id < b.id;
+  };
+  std::sort(existing.begin(), existing.end(), byID);
+  std::sort(entries.begin(), entries.end(), byID);
+  for (size_t i = 0; i < existing.size(); i++) {
+    if (existing[i].address != entries[i].address ||
+        existing[i].port != entries[i].port)
+      return true;
+  }
+  return false;
+}
+
+// ---------------- ServiceDefinition ----------------
+
+void ServiceDefinition::sendHeartbeat() {
+  registerService("passing");
+  std::string checkID = "service:" + id;
+  consul->updateTTL(checkID, "ok", "pass", [](bool ok, const std::string& err) {
+    if (!ok) LOG_WARN("service update TTL failed: %s", err.c_str());
+  });
+}
+
+void ServiceDefinition::registerWithInitialStatus() {
+  if (wasRegistered) return;
+  std::string status;
+  if (initialStatus == "passing") status = "passing";
+  else if (initialStatus == "warning") status = "warning";
+  else if (initialStatus == "critical") status = "critical";
+  LOG_INFO("Registering service %s with initial status set to %s",
+           name.c_str(), initialStatus.c_str());
+  registerService(status);
+}
+
+void ServiceDefinition::registerService(const std::string& status) {
+  if (wasRegistered || registerInFlight) return;
+  registerInFlight = true;
+  consul->serviceRegister(
+      id, name, tags, port, ipAddress, enableTagOverride, ttl, status,
+      deregisterCriticalServiceAfter, [this](bool ok, const std::string& err) {
+        registerInFlight = false;
+        if (!ok) {
+          LOG_WARN("service registration failed: %s", err.c_str());
+          return;
+        }
+        LOG_INFO("Service registered: %s", name.c_str());
+        wasRegistered = true;
+      });
+}
+
+void ServiceDefinition::deregister() {
+  LOG_DEBUG("deregistering: %s", id.c_str());
+  consul->serviceDeregister(id, [](bool ok, const std::string& err) {
+    if (!ok) LOG_INFO("deregistering failed: %s", err.c_str());
+  });
+}
+
+void ServiceDefinition::markForMaintenance() { deregister(); }
+
+}  // namespace cpilot

@@ -1,0 +1,141 @@
+#include "cpilot/events.hpp"
+
+#include <algorithm>
+
+#include "cpilot/log.hpp"
+
+namespace cpilot {
+
+const Event GlobalStartup{EventCode::Startup, "global"};
+const Event GlobalShutdown{EventCode::Shutdown, "global"};
+const Event NonEvent{EventCode::None, ""};
+const Event GlobalEnterMaintenance{EventCode::EnterMaintenance, "global"};
+const Event GlobalExitMaintenance{EventCode::ExitMaintenance, "global"};
+const Event QuitByTest{EventCode::Quit, "closed"};
+
+const char* eventCodeString(EventCode code) {
+  switch (code) {
+    case EventCode::None: return "None";
+    case EventCode::ExitSuccess: return "ExitSuccess";
+    case EventCode::ExitFailed: return "ExitFailed";
+    case EventCode::Stopping: return "Stopping";
+    case EventCode::Stopped: return "Stopped";
+    case EventCode::StatusHealthy: return "StatusHealthy";
+    case EventCode::StatusUnhealthy: return "StatusUnhealthy";
+    case EventCode::StatusChanged: return "StatusChanged";
+    case EventCode::TimerExpired: return "TimerExpired";
+    case EventCode::EnterMaintenance: return "EnterMaintenance";
+    case EventCode::ExitMaintenance: return "ExitMaintenance";
+    case EventCode::Error: return "Error";
+    case EventCode::Quit: return "Quit";
+    case EventCode::Metric: return "Metric";
+    case EventCode::Startup: return "Startup";
+    case EventCode::Shutdown: return "Shutdown";
+    case EventCode::Signal: return "Signal";
+  }
+  return "None";
+}
+
+bool eventCodeFromString(const std::string& name, EventCode* out) {
+  if (name == "exitSuccess") *out = EventCode::ExitSuccess;
+  else if (name == "exitFailed") *out = EventCode::ExitFailed;
+  else if (name == "stopping") *out = EventCode::Stopping;
+  else if (name == "stopped") *out = EventCode::Stopped;
+  else if (name == "healthy") *out = EventCode::StatusHealthy;
+  else if (name == "unhealthy") *out = EventCode::StatusUnhealthy;
+  else if (name == "changed") *out = EventCode::StatusChanged;
+  else if (name == "timerExpired") *out = EventCode::TimerExpired;
+  else if (name == "enterMaintenance") *out = EventCode::EnterMaintenance;
+  else if (name == "exitMaintenance") *out = EventCode::ExitMaintenance;
+  else if (name == "error") *out = EventCode::Error;
+  else if (name == "quit") *out = EventCode::Quit;
+  else if (name == "startup") *out = EventCode::Startup;
+  else if (name == "shutdown") *out = EventCode::Shutdown;
+  else if (name == "SIGHUP" || name == "SIGUSR2") *out = EventCode::Signal;
+  else return false;
+  return true;
+}
+
+Bus::Bus(Loop& loop) : loop_(loop), ring_(10) {
+  eventsCounter_ = prom::Registry::global().registerFamily(
+      "containerpilot_events",
+      "count of ContainerPilot events, partitioned by type and source",
+      prom::MetricType::Counter, {"code", "source"});
+  dispatchHist_ = prom::Registry::global().registerFamily(
+      "containerpilot_event_dispatch_seconds",
+      "latency from event publish to delivery", prom::MetricType::Histogram);
+}
+
+void Bus::subscribe(Subscriber* s) { subscribers_.push_back(s); }
+
+void Bus::unsubscribe(Subscriber* s) {
+  subscribers_.erase(std::remove(subscribers_.begin(), subscribers_.end(), s),
+                     subscribers_.end());
+}
+
+void Bus::publish(Event event) {
+  LOG_DEBUG("event: %s", event.str().c_str());
+  if (event.code != EventCode::Metric) {
+    eventsCounter_->inc({eventCodeString(event.code), event.source});
+  }
+  published_++;
+  // debug ring (events/bus.go:24-31)
+  ring_[(head_ + 1) % (int)ring_.size()] = event;
+  int old = head_;
+  head_ = (head_ + 1) % (int)ring_.size();
+  if (old != -1 && head_ == tail_) tail_ = (tail_ + 1) % (int)ring_.size();
+
+  queue_.emplace_back(std::move(event), Clock::now());
+  if (!drainScheduled_) {
+    drainScheduled_ = true;
+    loop_.defer([this] { drain(); });
+  }
+}
+
+void Bus::drain() {
+  drainScheduled_ = false;
+  // process what's queued now; publishes from handlers go to the next batch
+  size_t n = queue_.size();
+  for (size_t i = 0; i < n && !queue_.empty(); i++) {
+    auto [event, publishedAt] = std::move(queue_.front());
+    queue_.pop_front();
+    double latency =
+        std::chrono::duration<double>(Clock::now() - publishedAt).count();
+    dispatchHist_->observe(latency);
+    if (latencyWindow_.size() < latencyCap_) latencyWindow_.push_back(latency);
+    // snapshot: handlers may (un)subscribe during delivery. A subscriber
+    // removed mid-delivery must not receive the event (the reference
+    // panics-and-recovers for this; we make it impossible instead).
+    std::vector<Subscriber*> snapshot = subscribers_;
+    for (Subscriber* s : snapshot) {
+      if (std::find(subscribers_.begin(), subscribers_.end(), s) ==
+          subscribers_.end())
+        continue;
+      delivered_++;
+      s->onEvent(event);
+    }
+  }
+  if (!queue_.empty() && !drainScheduled_) {
+    drainScheduled_ = true;
+    loop_.defer([this] { drain(); });
+  }
+}
+
+std::vector<Event> Bus::debugEvents() {
+  std::vector<Event> out;
+  while (true) {
+    if (head_ == -1) break;
+    Event event = ring_[tail_ % ring_.size()];
+    if (tail_ == head_) {
+      head_ = -1;
+      tail_ = 0;
+    } else {
+      tail_ = (tail_ + 1) % (int)ring_.size();
+    }
+    if (event == NonEvent) break;
+    out.push_back(event);
+  }
+  return out;
+}
+
+}  // namespace cpilot

@@ -1,0 +1,377 @@
+#include "cpilot/http.hpp"
+
+#include <sys/epoll.h>
+
+#include <arpa/inet.h>
+#include <fcntl.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <poll.h>
+#include <sys/socket.h>
+#include <sys/un.h>
+#include <unistd.h>
+
+#include <cctype>
+#include <cstring>
+
+#include "cpilot/log.hpp"
+
+namespace cpilot {
+namespace http {
+
+const char* statusText(int code) {
+  switch (code) {
+    case 200: return "OK";
+    case 400: return "Bad Request";
+    case 404: return "Not Found";
+    case 405: return "Method Not Allowed";
+    case 422: return "Unprocessable Entity";
+    case 500: return "Internal Server Error";
+    default: return "";
+  }
+}
+
+struct Server::Conn {
+  int fd = -1;
+  std::string inbuf;
+  bool headersDone = false;
+  size_t contentLength = 0;
+  size_t headerEnd = 0;
+  Request req;
+};
+
+Server::Server(Loop& loop, Handler handler)
+    : loop_(loop), handler_(std::move(handler)) {}
+
+Server::~Server() { stop(); }
+
+static void setNonblock(int fd) {
+  fcntl(fd, F_SETFL, fcntl(fd, F_GETFL, 0) | O_NONBLOCK);
+  fcntl(fd, F_SETFD, FD_CLOEXEC);
+}
+
+bool Server::listenUnix(const std::string& path, std::string* err) {
+  int fd = socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) {
+    *err = strerror(errno);
+    return false;
+  }
+  struct sockaddr_un addr;
+  memset(&addr, 0, sizeof(addr));
+  addr.sun_family = AF_UNIX;
+  strncpy(addr.sun_path, path.c_str(), sizeof(addr.sun_path) - 1);
+  if (bind(fd, (struct sockaddr*)&addr, sizeof(addr)) != 0 ||
+      listen(fd, 128) != 0) {
+    *err = strerror(errno);
+    close(fd);
+    return false;
+  }
+  setNonblock(fd);
+  listenFd_ = fd;
+  loop_.watchFd(fd, EPOLLIN, [this](uint32_t) { acceptReady(); });
+  return true;
+}
+
+bool Server::listenTcp(const std::string& ip, int port, std::string* err) {
+  int fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) {
+    *err = strerror(errno);
+    return false;
+  }
+  int one = 1;
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  struct sockaddr_in addr;
+  memset(&addr, 0, sizeof(addr));
+  addr.sin_family = AF_INET;
+  addr.sin_port = htons(port);
+  if (ip.empty() || ip == "0.0.0.0") {
+    addr.sin_addr.s_addr = INADDR_ANY;
+  } else if (inet_pton(AF_INET, ip.c_str(), &addr.sin_addr) != 1) {
+    addr.sin_addr.s_addr = INADDR_ANY;
+  }
+  if (bind(fd, (struct sockaddr*)&addr, sizeof(addr)) != 0 ||
+      listen(fd, 128) != 0) {
+    *err = strerror(errno);
+    close(fd);
+    return false;
+  }
+  setNonblock(fd);
+  listenFd_ = fd;
+  loop_.watchFd(fd, EPOLLIN, [this](uint32_t) { acceptReady(); });
+  return true;
+}
+
+void Server::stop() {
+  if (listenFd_ >= 0) {
+    loop_.unwatchFd(listenFd_);
+    close(listenFd_);
+    listenFd_ = -1;
+  }
+  for (auto& kv : conns_) {
+    loop_.unwatchFd(kv.first);
+    close(kv.first);
+  }
+  conns_.clear();
+}
+
+void Server::acceptReady() {
+  while (true) {
+    int fd = accept(listenFd_, nullptr, nullptr);
+    if (fd < 0) return;
+    setNonblock(fd);
+    auto c = std::make_shared<Conn>();
+    c->fd = fd;
+    conns_[fd] = c;
+    loop_.watchFd(fd, EPOLLIN | EPOLLHUP,
+                  [this, c](uint32_t) { connReadable(c); });
+  }
+}
+
+void Server::closeConn(const std::shared_ptr<Conn>& c) {
+  if (c->fd < 0) return;
+  loop_.unwatchFd(c->fd);
+  close(c->fd);
+  conns_.erase(c->fd);
+  c->fd = -1;
+}
+
+static bool parseHeaders(Server::Conn* c);
+
+void Server::connReadable(std::shared_ptr<Conn> c) {
+  char buf[8192];
+  while (true) {
+    ssize_t n = read(c->fd, buf, sizeof(buf));
+    if (n > 0) {
+      c->inbuf.append(buf, n);
+      if (c->inbuf.size() > (1 << 22)) {  // 4 MiB cap
+        closeConn(c);
+        return;
+      }
+      continue;
+    }
+    if (n < 0 && errno == EAGAIN) break;
+    if (n < 0 && errno == EINTR) continue;
+    // EOF or error before a complete request
+    if (!c->headersDone || c->inbuf.size() < c->headerEnd + c->contentLength) {
+      closeConn(c);
+      return;
+    }
+    break;
+  }
+
+  if (!c->headersDone) {
+    size_t end = c->inbuf.find("\r\n\r\n");
+    if (end == std::string::npos) return;  // wait for more
+    c->headerEnd = end + 4;
+    if (!parseHeaders(c.get())) {
+      closeConn(c);
+      return;
+    }
+    c->headersDone = true;
+  }
+  if (c->inbuf.size() < c->headerEnd + c->contentLength) return;
+  c->req.body = c->inbuf.substr(c->headerEnd, c->contentLength);
+
+  Response resp = handler_(c->req);
+  std::string out = "HTTP/1.1 " + std::to_string(resp.status) + " " +
+                    statusText(resp.status) + "\r\n";
+  out += "Content-Type: " + resp.contentType + "\r\n";
+  out += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
+  out += "Connection: close\r\n\r\n";
+  out += resp.body;
+  size_t off = 0;
+  while (off < out.size()) {
+    ssize_t n = write(c->fd, out.data() + off, out.size() - off);
+    if (n > 0) {
+      off += n;
+    } else if (n < 0 && (errno == EAGAIN || errno == EINTR)) {
+      // short blocking wait; responses are small
+      struct pollfd pfd{c->fd, POLLOUT, 0};
+      poll(&pfd, 1, 1000);
+    } else {
+      break;
+    }
+  }
+  closeConn(c);
+}
+
+static bool parseHeaders(Server::Conn* c) {
+  const std::string& b = c->inbuf;
+  size_t lineEnd = b.find("\r\n");
+  if (lineEnd == std::string::npos) return false;
+  std::string reqLine = b.substr(0, lineEnd);
+  size_t sp1 = reqLine.find(' ');
+  size_t sp2 = reqLine.rfind(' ');
+  if (sp1 == std::string::npos || sp2 == sp1) return false;
+  c->req.method = reqLine.substr(0, sp1);
+  std::string target = reqLine.substr(sp1 + 1, sp2 - sp1 - 1);
+  size_t q = target.find('?');
+  if (q != std::string::npos) {
+    c->req.path = target.substr(0, q);
+    c->req.query = target.substr(q + 1);
+  } else {
+    c->req.path = target;
+  }
+  size_t pos = lineEnd + 2;
+  while (pos < c->headerEnd - 2) {
+    size_t eol = b.find("\r\n", pos);
+    if (eol == std::string::npos || eol >= c->headerEnd - 2) break;
+    std::string line = b.substr(pos, eol - pos);
+    pos = eol + 2;
+    size_t colon = line.find(':');
+    if (colon == std::string::npos) continue;
+    std::string key = line.substr(0, colon);
+    for (auto& ch : key) ch = tolower((unsigned char)ch);
+    size_t vstart = colon + 1;
+    while (vstart < line.size() && line[vstart] == ' ') vstart++;
+    c->req.headers[key] = line.substr(vstart);
+  }
+  auto it = c->req.headers.find("content-length");
+  c->contentLength = (it != c->req.headers.end())
+                         ? (size_t)atoll(it->second.c_str())
+                         : 0;
+  return true;
+}
+
+// ---------------- blocking client ----------------
+
+namespace {
+
+int connectTarget(const std::string& target, int timeoutMs, std::string* err) {
+  int fd = -1;
+  if (target.rfind("unix:", 0) == 0) {
+    fd = socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    if (fd < 0) {
+      *err = strerror(errno);
+      return -1;
+    }
+    struct sockaddr_un addr;
+    memset(&addr, 0, sizeof(addr));
+    addr.sun_family = AF_UNIX;
+    strncpy(addr.sun_path, target.c_str() + 5, sizeof(addr.sun_path) - 1);
+    if (connect(fd, (struct sockaddr*)&addr, sizeof(addr)) != 0) {
+      *err = strerror(errno);
+      close(fd);
+      return -1;
+    }
+  } else {
+    std::string host = target;
+    std::string port = "80";
+    size_t colon = target.rfind(':');
+    if (colon != std::string::npos) {
+      host = target.substr(0, colon);
+      port = target.substr(colon + 1);
+    }
+    struct addrinfo hints;
+    memset(&hints, 0, sizeof(hints));
+    hints.ai_family = AF_UNSPEC;
+    hints.ai_socktype = SOCK_STREAM;
+    struct addrinfo* res = nullptr;
+    int rc = getaddrinfo(host.c_str(), port.c_str(), &hints, &res);
+    if (rc != 0) {
+      *err = gai_strerror(rc);
+      return -1;
+    }
+    for (struct addrinfo* ai = res; ai; ai = ai->ai_next) {
+      fd = socket(ai->ai_family, ai->ai_socktype | SOCK_CLOEXEC,
+                  ai->ai_protocol);
+      if (fd < 0) continue;
+      if (connect(fd, ai->ai_addr, ai->ai_addrlen) == 0) break;
+      close(fd);
+      fd = -1;
+    }
+    freeaddrinfo(res);
+    if (fd < 0) {
+      *err = "connection refused";
+      return -1;
+    }
+  }
+  struct timeval tv;
+  tv.tv_sec = timeoutMs / 1000;
+  tv.tv_usec = (timeoutMs % 1000) * 1000;
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
+  return fd;
+}
+
+}  // namespace
+
+ClientResult request(const std::string& target, const std::string& method,
+                     const std::string& path, const std::string& body,
+                     const std::string& contentType,
+                     const std::map<std::string, std::string>& headers,
+                     int timeoutMs) {
+  ClientResult result;
+  int fd = connectTarget(target, timeoutMs, &result.error);
+  if (fd < 0) return result;
+
+  std::string host = target.rfind("unix:", 0) == 0 ? "localhost" : target;
+  std::string req = method + " " + path + " HTTP/1.1\r\n";
+  req += "Host: " + host + "\r\n";
+  req += "Connection: close\r\n";
+  for (auto& kv : headers) req += kv.first + ": " + kv.second + "\r\n";
+  if (!body.empty() || method == "POST" || method == "PUT") {
+    req += "Content-Type: " + contentType + "\r\n";
+    req += "Content-Length: " + std::to_string(body.size()) + "\r\n";
+  }
+  req += "\r\n";
+  req += body;
+
+  size_t off = 0;
+  while (off < req.size()) {
+    ssize_t n = write(fd, req.data() + off, req.size() - off);
+    if (n <= 0) {
+      if (n < 0 && errno == EINTR) continue;
+      result.error = "write failed";
+      close(fd);
+      return result;
+    }
+    off += n;
+  }
+
+  std::string resp;
+  char buf[8192];
+  while (true) {
+    ssize_t n = read(fd, buf, sizeof(buf));
+    if (n > 0) {
+      resp.append(buf, n);
+    } else if (n < 0 && errno == EINTR) {
+      continue;
+    } else {
+      break;
+    }
+  }
+  close(fd);
+
+  size_t headerEnd = resp.find("\r\n\r\n");
+  if (headerEnd == std::string::npos || resp.compare(0, 5, "HTTP/") != 0) {
+    result.error = "malformed response";
+    return result;
+  }
+  size_t sp = resp.find(' ');
+  result.status = atoi(resp.c_str() + sp + 1);
+  result.body = resp.substr(headerEnd + 4);
+  // chunked responses: dechunk (Consul uses Content-Length, but be safe)
+  std::string lower = resp.substr(0, headerEnd);
+  for (auto& ch : lower) ch = tolower((unsigned char)ch);
+  if (lower.find("transfer-encoding: chunked") != std::string::npos) {
+    std::string out;
+    size_t pos = 0;
+    const std::string& cb = result.body;
+    while (pos < cb.size()) {
+      size_t eol = cb.find("\r\n", pos);
+      if (eol == std::string::npos) break;
+      long len = strtol(cb.substr(pos, eol - pos).c_str(), nullptr, 16);
+      if (len <= 0) break;
+      out += cb.substr(eol + 2, len);
+      pos = eol + 2 + len + 2;
+    }
+    result.body = out;
+  }
+  result.ok = true;
+  return result;
+}
+
+}  // namespace http
+}  // namespace cpilot

@@ -1,0 +1,219 @@
+#include "cpilot/loop.hpp"
+
+#include <fcntl.h>
+#include <sys/epoll.h>
+#include <sys/timerfd.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <stdexcept>
+
+#include "cpilot/log.hpp"
+
+namespace cpilot {
+
+Loop::Loop() {
+  epfd_ = epoll_create1(EPOLL_CLOEXEC);
+  if (epfd_ < 0) throw std::runtime_error("epoll_create1 failed");
+  timerfd_ = timerfd_create(CLOCK_MONOTONIC, TFD_NONBLOCK | TFD_CLOEXEC);
+  if (timerfd_ < 0) throw std::runtime_error("timerfd_create failed");
+  if (pipe2(wakeupFds_, O_NONBLOCK | O_CLOEXEC) != 0)
+    throw std::runtime_error("pipe2 failed");
+
+  watchFd(timerfd_, EPOLLIN, [this](uint32_t) {
+    uint64_t expirations;
+    while (read(timerfd_, &expirations, sizeof(expirations)) > 0) {
+    }
+    fireDueTimers();
+  });
+  watchFd(wakeupFds_[0], EPOLLIN, [this](uint32_t) {
+    char buf[256];
+    while (read(wakeupFds_[0], buf, sizeof(buf)) > 0) {
+    }
+    drainPosted();
+  });
+}
+
+Loop::~Loop() {
+  if (epfd_ >= 0) close(epfd_);
+  if (timerfd_ >= 0) close(timerfd_);
+  if (wakeupFds_[0] >= 0) close(wakeupFds_[0]);
+  if (wakeupFds_[1] >= 0) close(wakeupFds_[1]);
+}
+
+void Loop::watchFd(int fd, uint32_t events, FdCallback cb) {
+  struct epoll_event ev;
+  memset(&ev, 0, sizeof(ev));
+  ev.events = events;
+  ev.data.fd = fd;
+  if (epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev) != 0)
+    throw std::runtime_error(std::string("epoll_ctl ADD failed: ") +
+                             strerror(errno));
+  fdCallbacks_[fd] = std::move(cb);
+}
+
+void Loop::modifyFd(int fd, uint32_t events) {
+  struct epoll_event ev;
+  memset(&ev, 0, sizeof(ev));
+  ev.events = events;
+  ev.data.fd = fd;
+  epoll_ctl(epfd_, EPOLL_CTL_MOD, fd, &ev);
+}
+
+void Loop::unwatchFd(int fd) {
+  epoll_ctl(epfd_, EPOLL_CTL_DEL, fd, nullptr);
+  fdCallbacks_.erase(fd);
+}
+
+uint64_t Loop::addTimeout(Ns delay, TimerCallback cb) {
+  auto t = std::make_shared<Timer>();
+  t->id = nextTimerId_++;
+  t->deadline = Clock::now() + delay;
+  t->interval = Ns(0);
+  t->cb = std::move(cb);
+  timersById_[t->id] = t;
+  timers_.push(t);
+  armTimerFd();
+  return t->id;
+}
+
+uint64_t Loop::addInterval(Ns interval, TimerCallback cb) {
+  if (interval <= Ns(0)) interval = Ns(1);
+  auto t = std::make_shared<Timer>();
+  t->id = nextTimerId_++;
+  t->deadline = Clock::now() + interval;
+  t->interval = interval;
+  t->cb = std::move(cb);
+  timersById_[t->id] = t;
+  timers_.push(t);
+  armTimerFd();
+  return t->id;
+}
+
+void Loop::cancelTimer(uint64_t id) {
+  auto it = timersById_.find(id);
+  if (it == timersById_.end()) return;
+  it->second->canceled = true;
+  timersById_.erase(it);
+}
+
+void Loop::defer(std::function<void()> fn) { deferred_.push_back(std::move(fn)); }
+
+void Loop::post(std::function<void()> fn) {
+  {
+    std::lock_guard<std::mutex> l(postedMu_);
+    posted_.push_back(std::move(fn));
+  }
+  char b = 1;
+  ssize_t unused = write(wakeupFds_[1], &b, 1);
+  (void)unused;
+}
+
+void Loop::watchChild(pid_t pid, ChildCallback cb) {
+  children_[pid] = std::move(cb);
+}
+
+void Loop::reapChildren() {
+  while (true) {
+    int status = 0;
+    pid_t pid = waitpid(-1, &status, WNOHANG);
+    if (pid <= 0) break;
+    auto it = children_.find(pid);
+    if (it != children_.end()) {
+      ChildCallback cb = std::move(it->second);
+      children_.erase(it);
+      cb(status);
+    }
+    // unknown pids: children reparented to us are reaped silently
+    // (only happens if we run as PID 1 without the sup split)
+  }
+}
+
+void Loop::armTimerFd() {
+  // pop canceled timers off the top
+  while (!timers_.empty() && timers_.top()->canceled) timers_.pop();
+  struct itimerspec its;
+  memset(&its, 0, sizeof(its));
+  if (!timers_.empty()) {
+    auto now = Clock::now();
+    Ns delta = std::chrono::duration_cast<Ns>(timers_.top()->deadline - now);
+    if (delta < Ns(1)) delta = Ns(1);
+    its.it_value.tv_sec = delta.count() / 1000000000LL;
+    its.it_value.tv_nsec = delta.count() % 1000000000LL;
+  }
+  timerfd_settime(timerfd_, 0, &its, nullptr);
+}
+
+void Loop::fireDueTimers() {
+  auto now = Clock::now();
+  while (!timers_.empty()) {
+    auto t = timers_.top();
+    if (t->canceled) {
+      timers_.pop();
+      continue;
+    }
+    if (t->deadline > now) break;
+    timers_.pop();
+    if (t->interval > Ns(0)) {
+      t->deadline = now + t->interval;
+      timers_.push(t);
+      t->cb();
+    } else {
+      timersById_.erase(t->id);
+      t->cb();
+    }
+  }
+  armTimerFd();
+}
+
+void Loop::drainDeferred() {
+  // bounded: only run what's queued now; callbacks may defer more which
+  // runs next iteration (prevents starvation of fds/timers)
+  size_t n = deferred_.size();
+  for (size_t i = 0; i < n && !deferred_.empty(); i++) {
+    auto fn = std::move(deferred_.front());
+    deferred_.pop_front();
+    fn();
+  }
+}
+
+void Loop::drainPosted() {
+  std::deque<std::function<void()>> batch;
+  {
+    std::lock_guard<std::mutex> l(postedMu_);
+    batch.swap(posted_);
+  }
+  for (auto& fn : batch) fn();
+}
+
+void Loop::run() {
+  stopped_ = false;
+  std::vector<struct epoll_event> events(64);
+  while (!stopped_) {
+    drainDeferred();
+    if (stopped_) break;
+    armTimerFd();
+    int timeoutMs = deferred_.empty() ? 1000 : 0;
+    int n = epoll_wait(epfd_, events.data(), (int)events.size(), timeoutMs);
+    if (n < 0) {
+      if (errno == EINTR) continue;
+      LOG_ERROR("epoll_wait: %s", strerror(errno));
+      break;
+    }
+    for (int i = 0; i < n && !stopped_; i++) {
+      int fd = events[i].data.fd;
+      auto it = fdCallbacks_.find(fd);
+      if (it != fdCallbacks_.end()) {
+        // copy: callback may unwatch itself
+        FdCallback cb = it->second;
+        cb(events[i].events);
+      }
+    }
+    fireDueTimers();
+  }
+}
+
+void Loop::stop() { stopped_ = true; }
+
+}  // namespace cpilot

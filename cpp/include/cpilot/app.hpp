@@ -52,6 +52,7 @@ class App {
   std::vector<std::shared_ptr<Watch>> watches_;
   int signalFd_ = -1;
   bool finishing_ = false;
+  uint64_t killSweepTimer_ = 0;
 
   // watches the bus for Shutdown so a generation with zero (or already
   // completed) jobs still finishes

@@ -72,7 +72,7 @@ class Subscriber {
   virtual void onEvent(const Event& event) = 0;
 };
 
-class Bus {
+class Bus : public std::enable_shared_from_this<Bus> {
  public:
   explicit Bus(Loop& loop);
 

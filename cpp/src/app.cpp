@@ -87,12 +87,29 @@ void App::setupSignals() {
 void App::ShutdownWatcher::onEvent(const Event& event) {
   if (event.code != EventCode::Shutdown) return;
   App* a = app;
+  // a generation with zero (or already completed) jobs still finishes
   a->loop_.defer([a] {
     for (auto& job : a->jobs_) {
       if (!job->isComplete()) return;
     }
     a->maybeFinishGeneration();
   });
+  // Bound the graceful drain: the reference relies on Docker's SIGKILL
+  // when a job's stop chain hangs past stopTimeout (core/app.go:146-157
+  // only sweeps after all jobs completed, so a hung pre-stop exec stalls
+  // shutdown forever). We SIGKILL all job process groups after
+  // stopTimeout so SIGTERM drains are always bounded.
+  if (a->killSweepTimer_ == 0 && a->cfg_->stopTimeout > 0) {
+    a->killSweepTimer_ = a->loop_.addTimeout(
+        std::chrono::seconds(a->cfg_->stopTimeout), [a] {
+          a->killSweepTimer_ = 0;
+          if (a->finishing_) return;
+          LOG_DEBUG("stop timeout exceeded, killing all job processes");
+          for (auto& job : a->jobs_) {
+            if (!job->isComplete()) job->kill();
+          }
+        });
+  }
 }
 
 void App::startGeneration() {
@@ -161,6 +178,10 @@ void App::maybeFinishGeneration() {
 }
 
 void App::teardownGeneration() {
+  if (killSweepTimer_) {
+    loop_.cancelTimer(killSweepTimer_);
+    killSweepTimer_ = 0;
+  }
   // aux components shut down after all jobs are complete
   // (core/app.go:104-140 completion watcher -> ctx cancel)
   if (control_) control_->stop();

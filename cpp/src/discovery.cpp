@@ -27,7 +27,8 @@ void parseRawURI(const std::string& raw, std::string* address,
 std::string urlEncode(const std::string& s) {
   std::string out;
   for (unsigned char c : s) {
-    if (isalnum(c) || c == '-' || c == '_' || c == '.' || c == '~') {
+    if (isalnum(c) || c == '-' || c == '_' || c == '.' || c == '~' ||
+        c == ':' || c == '@') {
       out += c;
     } else {
       char buf[4];

@@ -88,7 +88,9 @@ void Bus::publish(Event event) {
   queue_.emplace_back(std::move(event), Clock::now());
   if (!drainScheduled_) {
     drainScheduled_ = true;
-    loop_.defer([this] { drain(); });
+    // keep the bus alive until the drain runs: a reload can otherwise
+    // free this generation's bus with its drain still queued
+    loop_.defer([self = shared_from_this()] { self->drain(); });
   }
 }
 
@@ -117,7 +119,7 @@ void Bus::drain() {
   }
   if (!queue_.empty() && !drainScheduled_) {
     drainScheduled_ = true;
-    loop_.defer([this] { drain(); });
+    loop_.defer([self = shared_from_this()] { self->drain(); });
   }
 }
 

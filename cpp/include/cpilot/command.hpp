@@ -57,10 +57,12 @@ class Command : public std::enable_shared_from_this<Command> {
   Duration timeout_;
   bool raw_;              // raw log passthrough (no wrapping)
   std::string logField_;  // "job" or "check" label value for wrapped logs
+  std::string pidEnvName_;  // cached CONTAINERPILOT_{NAME}_PID
 
   pid_t pid_ = -1;
   bool running_ = false;
   bool pendingRun_ = false;
+  int pendingSignal_ = 0;  // term/kill requested while spawn in flight
   Loop* loop_ = nullptr;
   uint64_t timeoutTimer_ = 0;
   int logFd_ = -1;

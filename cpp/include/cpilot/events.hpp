@@ -104,7 +104,8 @@ class Bus : public std::enable_shared_from_this<Bus> {
   void drain();
 
   Loop& loop_;
-  std::vector<Subscriber*> subscribers_;
+  std::vector<Subscriber*> subscribers_;  // nullptr = tombstoned slot
+  bool tombstones_ = false;
   std::deque<std::pair<Event, TimePoint>> queue_;
   bool drainScheduled_ = false;
   bool reload_ = false;
@@ -116,6 +117,7 @@ class Bus : public std::enable_shared_from_this<Bus> {
   uint64_t published_ = 0, delivered_ = 0;
   std::shared_ptr<prom::Family> eventsCounter_;
   std::shared_ptr<prom::Family> dispatchHist_;
+  std::shared_ptr<prom::Family> deliveriesCounter_;
   std::vector<double> latencyWindow_;
   size_t latencyCap_ = 262144;
 };

@@ -124,6 +124,10 @@ class Job : public Subscriber, public std::enable_shared_from_this<Job> {
   void finishCleanup();
 
   std::string name_;
+  // precomputed event-match sources (hot path: every bus event hits
+  // dispatch() in every job)
+  std::string heartbeatSource_, runEverySource_, healthCheckName_,
+      stoppingTimeoutSource_;
   CommandPtr exec_;
   JobStatus status_ = JobStatus::Idle;
   std::shared_ptr<ServiceDefinition> service_;

@@ -45,7 +45,8 @@ class Loop {
 
   // --- timers (ids are never 0) ---
   uint64_t addTimeout(Ns delay, TimerCallback cb);
-  uint64_t addInterval(Ns interval, TimerCallback cb);
+  // initialDelay < 0 means "one full interval" (default phase)
+  uint64_t addInterval(Ns interval, TimerCallback cb, Ns initialDelay = Ns(-1));
   void cancelTimer(uint64_t id);
 
   // --- deferred work (runs on the next loop iteration, FIFO) ---
@@ -55,6 +56,9 @@ class Loop {
   void post(std::function<void()> fn);
 
   // --- children ---
+  // Registers interest in a child's exit. If the child was already
+  // reaped (spawn-completion can race SIGCHLD when spawning happens on
+  // the spawner thread), the stored status is delivered immediately.
   void watchChild(pid_t pid, ChildCallback cb);
   // call on SIGCHLD: waitpid(-1, WNOHANG) loop dispatching to callbacks
   void reapChildren();
@@ -102,6 +106,7 @@ class Loop {
   std::deque<std::function<void()>> posted_;
 
   std::map<pid_t, ChildCallback> children_;
+  std::map<pid_t, int> unclaimedExits_;
 };
 
 }  // namespace cpilot

@@ -24,6 +24,10 @@ struct HistogramData {
   uint64_t count = 0;
   double sum = 0;
   HistogramData() : counts(bounds.size(), 0) {}
+  void setBounds(std::vector<double> b) {
+    bounds = std::move(b);
+    counts.assign(bounds.size(), 0);
+  }
   void observe(double v) {
     for (size_t i = 0; i < bounds.size(); i++)
       if (v <= bounds[i]) counts[i]++;
@@ -86,6 +90,12 @@ class Family {
     else c.summ.observe(v);
   }
   void observe(double v) { observe({}, v); }
+  // set histogram bucket bounds (before first observe)
+  void setBuckets(std::vector<double> bounds) {
+    std::lock_guard<std::mutex> l(mu_);
+    bucketBounds_ = bounds;
+    for (auto& kv : children_) kv.second.hist.setBounds(bounds);
+  }
 
   void expose(std::string& out) const;
   const std::string& name() const { return name_; }
@@ -93,8 +103,13 @@ class Family {
 
  private:
   Child& child(const std::vector<std::string>& labels) {
-    return children_[labels];
+    auto it = children_.find(labels);
+    if (it != children_.end()) return it->second;
+    Child& c = children_[labels];
+    if (!bucketBounds_.empty()) c.hist.setBounds(bucketBounds_);
+    return c;
   }
+  std::vector<double> bucketBounds_;
   std::string name_, help_;
   MetricType type_;
   std::vector<std::string> labelNames_;

@@ -10,6 +10,7 @@
 #include <cstring>
 
 #include "cpilot/log.hpp"
+#include "cpilot/spawner.hpp"
 
 namespace cpilot {
 
@@ -118,95 +119,82 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
     }
   }
 
-  pid_t pid = fork();
-  if (pid < 0) {
-    LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(errno));
-    if (pipefds[0] >= 0) {
-      close(pipefds[0]);
-      close(pipefds[1]);
-    }
-    bus->publish(Event{EventCode::ExitFailed, name_});
-    bus->publish(Event{EventCode::Error, strerror(errno)});
-    return;
-  }
-  if (pid == 0) {
-    // child
-    setpgid(0, 0);
-    if (!raw_) {
-      dup2(pipefds[1], 1);
-      dup2(pipefds[1], 2);
-    }
-    // unblock all signals (the daemon blocks them for signalfd)
-    sigset_t empty;
-    sigemptyset(&empty);
-    sigprocmask(SIG_SETMASK, &empty, nullptr);
-    std::vector<char*> argv;
-    argv.push_back(const_cast<char*>(exec_.c_str()));
-    for (auto& a : args_) argv.push_back(const_cast<char*>(a.c_str()));
-    argv.push_back(nullptr);
-    execvp(exec_.c_str(), argv.data());
-    fprintf(stderr, "unable to start %s: %s\n", exec_.c_str(),
-            strerror(errno));
-    _exit(127);
-  }
-
-  // parent
-  setpgid(pid, pid);  // avoid race with child's setpgid
-  pid_ = pid;
   running_ = true;
-
-  std::string pidEnv = "CONTAINERPILOT_" + envName() + "_PID";
-  setenv(pidEnv.c_str(), std::to_string(pid).c_str(), 1);
-
-  if (!raw_) {
-    close(pipefds[1]);
-    logFd_ = pipefds[0];
-    fcntl(logFd_, F_SETFL, O_NONBLOCK);
-    auto self = shared_from_this();
-    loop.watchFd(logFd_, EPOLLIN | EPOLLHUP, [this, self](uint32_t) {
-      char buf[4096];
-      while (true) {
-        ssize_t n = read(logFd_, buf, sizeof(buf));
-        if (n > 0) {
-          logBuf_.append(buf, n);
-          size_t pos;
-          while ((pos = logBuf_.find('\n')) != std::string::npos) {
-            logging::logFields(logging::Level::Info, logField_, pid_,
-                               logBuf_.substr(0, pos));
-            logBuf_.erase(0, pos + 1);
-          }
-        } else if (n == 0 || (n < 0 && errno != EAGAIN)) {
-          if (!logBuf_.empty()) {
-            logging::logFields(logging::Level::Info, logField_, pid_, logBuf_);
-            logBuf_.clear();
-          }
-          loop_->unwatchFd(logFd_);
-          close(logFd_);
-          logFd_ = -1;
-          return;
-        } else {
-          return;  // EAGAIN
-        }
-      }
-    });
-  }
-
-  if (timeout_ > Duration(0)) {
-    auto self = shared_from_this();
-    timeoutTimer_ = loop.addTimeout(timeout_, [this, self] {
-      LOG_WARN("%s timeout after %llds", name_.c_str(),
-               (long long)std::chrono::duration_cast<std::chrono::seconds>(
-                   timeout_)
-                   .count());
-      timeoutTimer_ = 0;
-      kill();
-    });
-  }
-
+  pid_ = -1;
+  pendingSignal_ = 0;
   auto self = shared_from_this();
-  loop.watchChild(pid, [this, self, bus](int status) {
-    onExit(*loop_, bus, status);
-  });
+  int readFd = pipefds[0];
+  // the spawner thread does the posix_spawnp so a burst of launches never
+  // blocks event dispatch; completion lands back on the loop
+  Spawner::global().spawn(
+      loop, exec_, args_, raw_ ? -1 : pipefds[1],
+      [this, self, bus, readFd](pid_t pid, int err) {
+        if (pid < 0) {
+          LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(err));
+          if (readFd >= 0) close(readFd);
+          running_ = false;
+          bus->publish(Event{EventCode::ExitFailed, name_});
+          bus->publish(Event{EventCode::Error, strerror(err)});
+          return;
+        }
+        pid_ = pid;
+        if (pidEnvName_.empty())
+          pidEnvName_ = "CONTAINERPILOT_" + envName() + "_PID";
+        setenv(pidEnvName_.c_str(), std::to_string(pid).c_str(), 1);
+
+        if (!raw_) {
+          logFd_ = readFd;
+          fcntl(logFd_, F_SETFL, O_NONBLOCK);
+          loop_->watchFd(logFd_, EPOLLIN | EPOLLHUP, [this, self](uint32_t) {
+            char buf[4096];
+            while (true) {
+              ssize_t n = read(logFd_, buf, sizeof(buf));
+              if (n > 0) {
+                logBuf_.append(buf, n);
+                size_t pos;
+                while ((pos = logBuf_.find('\n')) != std::string::npos) {
+                  logging::logFields(logging::Level::Info, logField_, pid_,
+                                     logBuf_.substr(0, pos));
+                  logBuf_.erase(0, pos + 1);
+                }
+              } else if (n == 0 || (n < 0 && errno != EAGAIN)) {
+                if (!logBuf_.empty()) {
+                  logging::logFields(logging::Level::Info, logField_, pid_,
+                                     logBuf_);
+                  logBuf_.clear();
+                }
+                loop_->unwatchFd(logFd_);
+                close(logFd_);
+                logFd_ = -1;
+                return;
+              } else {
+                return;  // EAGAIN
+              }
+            }
+          });
+        }
+
+        if (timeout_ > Duration(0)) {
+          timeoutTimer_ = loop_->addTimeout(timeout_, [this, self] {
+            LOG_WARN("%s timeout after %llds", name_.c_str(),
+                     (long long)std::chrono::duration_cast<
+                         std::chrono::seconds>(timeout_)
+                         .count());
+            timeoutTimer_ = 0;
+            kill();
+          });
+        }
+
+        loop_->watchChild(pid, [this, self, bus](int status) {
+          onExit(*loop_, bus, status);
+        });
+
+        // a term/kill that arrived while the spawn was in flight
+        if (pendingSignal_ != 0 && pid_ > 0) {
+          ::kill(-pid_, pendingSignal_);
+          pendingSignal_ = 0;
+        }
+      });
 }
 
 void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
@@ -235,8 +223,7 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
     logFd_ = -1;
   }
 
-  std::string pidEnv = "CONTAINERPILOT_" + envName() + "_PID";
-  unsetenv(pidEnv.c_str());
+  unsetenv(pidEnvName_.c_str());
 
   running_ = false;
   pid_ = -1;
@@ -265,6 +252,8 @@ void Command::term() {
   if (pid_ > 0) {
     LOG_DEBUG("terminating command '%s' at pid: %d", name_.c_str(), pid_);
     ::kill(-pid_, SIGTERM);
+  } else if (running_ && pendingSignal_ != SIGKILL) {
+    pendingSignal_ = SIGTERM;  // spawn in flight
   }
 }
 
@@ -273,6 +262,8 @@ void Command::kill() {
   if (pid_ > 0) {
     LOG_DEBUG("killing command '%s' at pid: %d", name_.c_str(), pid_);
     ::kill(-pid_, SIGKILL);
+  } else if (running_) {
+    pendingSignal_ = SIGKILL;  // spawn in flight
   }
 }
 

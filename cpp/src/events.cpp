@@ -64,13 +64,27 @@ Bus::Bus(Loop& loop) : loop_(loop), ring_(10) {
   dispatchHist_ = prom::Registry::global().registerFamily(
       "containerpilot_event_dispatch_seconds",
       "latency from event publish to delivery", prom::MetricType::Histogram);
+  // sub-millisecond resolution: the perf target is <1ms p99 dispatch
+  dispatchHist_->setBuckets({1e-5, 2e-5, 5e-5, 1e-4, 2.5e-4, 5e-4, 1e-3,
+                             2.5e-3, 5e-3, 1e-2, 5e-2, 0.1});
+  deliveriesCounter_ = prom::Registry::global().registerFamily(
+      "containerpilot_event_deliveries",
+      "count of event deliveries to subscribers (bus fan-out dispatches)",
+      prom::MetricType::Counter);
 }
 
 void Bus::subscribe(Subscriber* s) { subscribers_.push_back(s); }
 
 void Bus::unsubscribe(Subscriber* s) {
-  subscribers_.erase(std::remove(subscribers_.begin(), subscribers_.end(), s),
-                     subscribers_.end());
+  // tombstone instead of erase: delivery checks the slot in O(1) rather
+  // than std::find-ing membership per delivery (which made dispatch
+  // O(subscribers^2) per event); slots compact after each drain batch
+  for (auto& slot : subscribers_) {
+    if (slot == s) {
+      slot = nullptr;
+      tombstones_ = true;
+    }
+  }
 }
 
 void Bus::publish(Event event) {
@@ -105,17 +119,25 @@ void Bus::drain() {
         std::chrono::duration<double>(Clock::now() - publishedAt).count();
     dispatchHist_->observe(latency);
     if (latencyWindow_.size() < latencyCap_) latencyWindow_.push_back(latency);
-    // snapshot: handlers may (un)subscribe during delivery. A subscriber
-    // removed mid-delivery must not receive the event (the reference
-    // panics-and-recovers for this; we make it impossible instead).
-    std::vector<Subscriber*> snapshot = subscribers_;
-    for (Subscriber* s : snapshot) {
-      if (std::find(subscribers_.begin(), subscribers_.end(), s) ==
-          subscribers_.end())
-        continue;
+    // handlers may (un)subscribe during delivery: unsubscribes tombstone
+    // their slot (checked per delivery), new subscribes append past the
+    // bound captured here so they don't see this event
+    size_t bound = subscribers_.size();
+    uint64_t batch = 0;
+    for (size_t si = 0; si < bound; si++) {
+      Subscriber* s = subscribers_[si];
+      if (!s) continue;
       delivered_++;
+      batch++;
       s->onEvent(event);
     }
+    if (batch) deliveriesCounter_->inc({}, (double)batch);
+  }
+  if (tombstones_) {
+    subscribers_.erase(
+        std::remove(subscribers_.begin(), subscribers_.end(), nullptr),
+        subscribers_.end());
+    tombstones_ = false;
   }
   if (!queue_.empty() && !drainScheduled_) {
     drainScheduled_ = true;

@@ -198,10 +198,26 @@ bool validateHealthCheck(const Json& raw, std::shared_ptr<JobConfig>& cfg,
     *err = "job configuration error: " + *err;
     return false;
   }
-  int heartbeat = 0, ttl = 0;
-  if (const Json* v = health->find("interval")) decode::toInt(*v, &heartbeat);
+  // interval: int = seconds (reference-compatible, jobs/config.go:74) or a
+  // Go duration string (extension, so sub-second check cadences like the
+  // baseline's 100ms are expressible)
+  Duration heartbeat{0};
+  int ttl = 0;
+  if (const Json* v = health->find("interval")) {
+    if (v->isString() && !v->str().empty() &&
+        !isdigit((unsigned char)v->str()[0]) ) {
+      *err = "job[" + cfg->name + "].health.interval must be > 0";
+      return false;
+    }
+    try {
+      heartbeat = parseDuration(*v);
+    } catch (const std::exception&) {
+      *err = "job[" + cfg->name + "].health.interval must be > 0";
+      return false;
+    }
+  }
   if (const Json* v = health->find("ttl")) decode::toInt(*v, &ttl);
-  if (heartbeat < 1) {
+  if (heartbeat < std::chrono::milliseconds(1)) {
     *err = "job[" + cfg->name + "].health.interval must be > 0";
     return false;
   }
@@ -210,7 +226,7 @@ bool validateHealthCheck(const Json& raw, std::shared_ptr<JobConfig>& cfg,
     return false;
   }
   cfg->ttl = ttl;
-  cfg->heartbeatInterval = std::chrono::seconds(heartbeat);
+  cfg->heartbeatInterval = heartbeat;
 
   Duration checkTimeout = cfg->heartbeatInterval;
   std::string timeoutStr;
@@ -464,6 +480,11 @@ Job::Job(const std::shared_ptr<JobConfig>& cfg)
       restartLimit_(cfg->restartLimit),
       restartsRemain_(cfg->restartLimit),
       frequency_(cfg->freqInterval) {
+  heartbeatSource_ = name_ + ".heartbeat";
+  runEverySource_ = name_ + ".run-every";
+  healthCheckName_ =
+      healthCheckExec_ ? healthCheckExec_->name() : ("check." + name_);
+  stoppingTimeoutSource_ = name_ + ".stopping-timeout";
   if (name_ == "containerpilot") {
     // hardcoded always-healthy telemetry job (jobs/jobs.go:82-87)
     status_ = JobStatus::AlwaysHealthy;
@@ -477,19 +498,30 @@ void Job::run(Loop& loop, std::shared_ptr<Bus> bus,
   completedCb_ = std::move(completedCb);
   auto self = shared_from_this();
 
+  // Stagger each job's periodic timers by a deterministic per-name phase
+  // inside the first interval. Without this every job's check fires in
+  // the same millisecond (timers are all created at GlobalStartup),
+  // producing spawn/exit storms that batch the bus and blow out dispatch
+  // latency; with it the load spreads evenly across the period.
+  auto phase = [this](Duration interval) {
+    uint64_t h = std::hash<std::string>{}(name_);
+    return Duration(interval.count() / 2 +
+                    (Ns::rep)(h % 1000) * interval.count() / 2000);
+  };
+
   if (frequency_ > Duration(0)) {
     freqTimer_ = loop.addInterval(frequency_, [this, self] {
       LOG_DEBUG("timer: {TimerExpired %s.run-every}", name_.c_str());
-      processEvent(Event{EventCode::TimerExpired, name_ + ".run-every"});
-    });
+      processEvent(Event{EventCode::TimerExpired, runEverySource_});
+    }, phase(frequency_));
   }
   if (heartbeat_ > Duration(0)) {
     heartbeatTimer_ = loop.addInterval(heartbeat_, [this, self] {
       // heartbeat ticks for the telemetry job are not logged [GH-556]
       if (name_ != "containerpilot")
         LOG_DEBUG("timer: {TimerExpired %s.heartbeat}", name_.c_str());
-      processEvent(Event{EventCode::TimerExpired, name_ + ".heartbeat"});
-    });
+      processEvent(Event{EventCode::TimerExpired, heartbeatSource_});
+    }, phase(heartbeat_));
   }
   if (startTimeout_ > Duration(0)) {
     std::string timeoutName = name_ + ".wait-timeout";
@@ -519,7 +551,8 @@ void Job::processEvent(const Event& event) {
     // cleanup wait loop: only the awaited Stopped event or the stopping
     // timeout break it (jobs/jobs.go:397-407)
     if (event == stoppingWaitEvent_ ||
-        event == Event{EventCode::TimerExpired, name_ + ".stopping-timeout"}) {
+        (event.code == EventCode::TimerExpired &&
+         event.source == stoppingTimeoutSource_)) {
       finishCleanup();
     }
     return;
@@ -533,32 +566,42 @@ void Job::processEvent(const Event& event) {
 }
 
 Job::HandleResult Job::dispatch(const Event& event) {
-  // match order mirrors the reference switch (jobs/jobs.go:195-232)
-  std::string heartbeatSource = name_ + ".heartbeat";
-  std::string runEverySource = name_ + ".run-every";
-  std::string healthCheckName =
-      healthCheckExec_ ? healthCheckExec_->name() : ("check." + name_);
-
-  if (event == Event{EventCode::TimerExpired, heartbeatSource})
-    return onHeartbeatTimerExpired();
-  if (startTimeoutEvent_ != NonEvent && event == startTimeoutEvent_)
-    return onStartTimeoutExpired();
-  if (event == Event{EventCode::TimerExpired, runEverySource})
-    return onRunEveryTimerExpired();
-  if (event == Event{EventCode::ExitFailed, healthCheckName})
-    return onHealthCheckFailed();
-  if (event == Event{EventCode::ExitSuccess, healthCheckName})
-    return onHealthCheckPassed();
-  if (event == Event{EventCode::Quit, name_} || event == GlobalShutdown)
-    return onQuit();
-  if (event == GlobalEnterMaintenance) return onEnterMaintenance();
-  if (event == GlobalExitMaintenance) return onExitMaintenance();
-  if (event == Event{EventCode::ExitSuccess, name_} ||
-      event == Event{EventCode::ExitFailed, name_})
-    return onExecExit();
-  if (event == Event{EventCode::Signal, "SIGHUP"} ||
-      event == Event{EventCode::Signal, "SIGUSR2"})
-    return onSignalEvent(event.source);
+  // match order mirrors the reference switch (jobs/jobs.go:195-232);
+  // code compared before source so non-matching events exit cheaply
+  switch (event.code) {
+    case EventCode::TimerExpired:
+      if (event.source == heartbeatSource_) return onHeartbeatTimerExpired();
+      if (startTimeoutEvent_ != NonEvent && event == startTimeoutEvent_)
+        return onStartTimeoutExpired();
+      if (event.source == runEverySource_) return onRunEveryTimerExpired();
+      break;
+    case EventCode::ExitFailed:
+      if (event.source == healthCheckName_) return onHealthCheckFailed();
+      if (event.source == name_) return onExecExit();
+      break;
+    case EventCode::ExitSuccess:
+      if (event.source == healthCheckName_) return onHealthCheckPassed();
+      if (event.source == name_) return onExecExit();
+      break;
+    case EventCode::Quit:
+      if (event.source == name_) return onQuit();
+      break;
+    case EventCode::Shutdown:
+      if (event == GlobalShutdown) return onQuit();
+      break;
+    case EventCode::EnterMaintenance:
+      if (event == GlobalEnterMaintenance) return onEnterMaintenance();
+      break;
+    case EventCode::ExitMaintenance:
+      if (event == GlobalExitMaintenance) return onExitMaintenance();
+      break;
+    case EventCode::Signal:
+      if (event.source == "SIGHUP" || event.source == "SIGUSR2")
+        return onSignalEvent(event.source);
+      break;
+    default:
+      break;
+  }
   if (event == startEvent_) return onStartEvent();
   return kContinue;
 }

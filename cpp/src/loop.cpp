@@ -78,11 +78,12 @@ uint64_t Loop::addTimeout(Ns delay, TimerCallback cb) {
   return t->id;
 }
 
-uint64_t Loop::addInterval(Ns interval, TimerCallback cb) {
+uint64_t Loop::addInterval(Ns interval, TimerCallback cb, Ns initialDelay) {
   if (interval <= Ns(0)) interval = Ns(1);
+  if (initialDelay < Ns(0)) initialDelay = interval;
   auto t = std::make_shared<Timer>();
   t->id = nextTimerId_++;
-  t->deadline = Clock::now() + interval;
+  t->deadline = Clock::now() + initialDelay;
   t->interval = interval;
   t->cb = std::move(cb);
   timersById_[t->id] = t;
@@ -111,6 +112,13 @@ void Loop::post(std::function<void()> fn) {
 }
 
 void Loop::watchChild(pid_t pid, ChildCallback cb) {
+  auto it = unclaimedExits_.find(pid);
+  if (it != unclaimedExits_.end()) {
+    int status = it->second;
+    unclaimedExits_.erase(it);
+    cb(status);
+    return;
+  }
   children_[pid] = std::move(cb);
 }
 
@@ -124,9 +132,13 @@ void Loop::reapChildren() {
       ChildCallback cb = std::move(it->second);
       children_.erase(it);
       cb(status);
+    } else {
+      // exit raced the spawner thread's completion post; stash it for
+      // the watchChild that is about to arrive. (A child we never
+      // watched is an inherited zombie; the map stays tiny because every
+      // spawn claims its exit.)
+      unclaimedExits_[pid] = status;
     }
-    // unknown pids: children reparented to us are reaped silently
-    // (only happens if we run as PID 1 without the sup split)
   }
 }
 
@@ -156,7 +168,16 @@ void Loop::fireDueTimers() {
     if (t->deadline > now) break;
     timers_.pop();
     if (t->interval > Ns(0)) {
-      t->deadline = now + t->interval;
+      // keep the original phase (deadline += interval, not now + interval):
+      // re-anchoring to `now` lets one slow iteration cluster every
+      // periodic timer into the same instant permanently
+      t->deadline += t->interval;
+      if (t->deadline <= now) {
+        // missed whole periods: skip forward, preserving phase
+        auto behind = std::chrono::duration_cast<Ns>(now - t->deadline);
+        auto periods = behind.count() / t->interval.count() + 1;
+        t->deadline += Ns(periods * t->interval.count());
+      }
       timers_.push(t);
       t->cb();
     } else {

@@ -28,6 +28,7 @@ class MockConsul:
         self.deregistered = []  # serviceIDs
         self.requests = []      # (method, path)
         self.health = {}        # service name -> list of dicts(ID,Address,Port)
+        self.tokens = []        # X-Consul-Token header values seen
 
         outer = self
 
@@ -48,6 +49,9 @@ class MockConsul:
                 parsed = urlparse(self.path)
                 with outer.lock:
                     outer.requests.append(("GET", parsed.path))
+                    tok = self.headers.get("X-Consul-Token")
+                    if tok:
+                        outer.tokens.append(tok)
                 if parsed.path.startswith("/v1/health/service/"):
                     name = parsed.path[len("/v1/health/service/"):]
                     qs = parse_qs(parsed.query)
@@ -78,6 +82,9 @@ class MockConsul:
                 payload = json.loads(raw) if raw else {}
                 with outer.lock:
                     outer.requests.append(("PUT", parsed.path))
+                    tok = self.headers.get("X-Consul-Token")
+                    if tok:
+                        outer.tokens.append(tok)
                     if parsed.path == "/v1/agent/service/register":
                         outer.services[payload.get("ID", "")] = payload
                         return self._respond(200)

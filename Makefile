@@ -1,0 +1,24 @@
+# Convenience wrapper around the CMake build (reference: makefile).
+BUILD_DIR := build
+
+.PHONY: all build test unit integration bench clean
+
+all: build
+
+build:
+	cmake -S . -B $(BUILD_DIR) -G Ninja -DCMAKE_BUILD_TYPE=Release
+	ninja -C $(BUILD_DIR)
+
+unit: build
+	./bin/cpilot_unittests
+
+integration: build
+	python3 -m pytest tests/ -q -m "not gpu"
+
+test: unit integration
+
+bench: build
+	python3 bench.py --steps 30 --warmup 5
+
+clean:
+	rm -rf $(BUILD_DIR) bin containerpilot_amd/_native.so

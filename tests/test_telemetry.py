@@ -154,3 +154,39 @@ def test_synthetic_containerpilot_service_registers(daemon_factory,
     d.terminate()
     assert d.wait(timeout=30) == 0
     assert cp_id in mock_consul.deregistered
+
+
+def test_histogram_and_summary_metrics(daemon_factory, mock_consul):
+    """histogram/summary collectors expose buckets/quantiles after
+    observations (telemetry/metrics_config.go:63-80)."""
+    port = free_port()
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "jobs": [{"name": "main-app", "exec": "sleep 60"}],
+        "telemetry": {
+            "port": port,
+            "interfaces": ["static:127.0.0.1"],
+            "metrics": [
+                {"namespace": "app", "subsystem": "rq", "name": "latency",
+                 "help": "h", "type": "histogram"},
+                {"namespace": "app", "subsystem": "rq", "name": "sizes",
+                 "help": "s", "type": "summary"},
+            ],
+        },
+    }).start()
+    d.wait_for_socket()
+    url = "http://127.0.0.1:%d/metrics" % port
+    assert wait_until(lambda: _up(url))
+    for v in (0.01, 0.02, 0.3, 2.0):
+        d.control("POST", "/v3/metric",
+                  json.dumps({"app_rq_latency": v, "app_rq_sizes": v * 100}))
+    assert wait_until(
+        lambda: "app_rq_latency_count 4" in http_get(url)[1])
+    _, body = http_get(url)
+    assert 'app_rq_latency_bucket{le="0.025"} 2' in body
+    assert "app_rq_latency_sum" in body
+    assert 'app_rq_sizes{quantile="0.5"}' in body
+    assert "app_rq_sizes_count 4" in body
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -22,3 +22,18 @@ bench: build
 
 clean:
 	rm -rf $(BUILD_DIR) bin containerpilot_amd/_native.so
+
+# Sanitizer builds (evidence in profiles/sanitizers.md)
+tsan:
+	cmake -S . -B build-tsan -G Ninja -DCMAKE_BUILD_TYPE=RelWithDebInfo \
+	  -DCMAKE_CXX_FLAGS="-fsanitize=thread -g -O1"
+	ninja -C build-tsan cpilot_unittests containerpilot
+	./bin/cpilot_unittests
+	ninja -C $(BUILD_DIR)  # restore release binaries
+
+asan:
+	cmake -S . -B build-asan -G Ninja -DCMAKE_BUILD_TYPE=RelWithDebInfo \
+	  -DCMAKE_CXX_FLAGS="-fsanitize=address,undefined -g -O1"
+	ninja -C build-asan cpilot_unittests containerpilot
+	./bin/cpilot_unittests
+	ninja -C $(BUILD_DIR)

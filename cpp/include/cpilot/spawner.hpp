@@ -26,9 +26,11 @@ class Spawner {
   static Spawner& global();
 
   // stdioFd >= 0 is dup2'd onto the child's stdout+stderr and closed in
-  // the parent after the spawn completes.
+  // the parent after the spawn completes. env is a snapshot of the
+  // environment taken ON THE LOOP THREAD — posix_spawn must never read
+  // the live environ while the reactor setenv()s (data race -> EFAULT).
   void spawn(Loop& loop, std::string execPath, std::vector<std::string> args,
-             int stdioFd, SpawnCb cb);
+             std::vector<std::string> env, int stdioFd, SpawnCb cb);
 
  private:
   Spawner();
@@ -38,6 +40,7 @@ class Spawner {
     Loop* loop;
     std::string execPath;
     std::vector<std::string> args;
+    std::vector<std::string> env;
     int stdioFd;
     SpawnCb cb;
   };

@@ -12,6 +12,8 @@
 #include "cpilot/log.hpp"
 #include "cpilot/spawner.hpp"
 
+extern char** environ;
+
 namespace cpilot {
 
 bool parseArgs(const Json& raw, std::string* execPath,
@@ -124,10 +126,14 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   pendingSignal_ = 0;
   auto self = shared_from_this();
   int readFd = pipefds[0];
+  // snapshot env on this (the loop) thread; the spawner must not read
+  // the live environ concurrently with our setenv calls
+  std::vector<std::string> envCopy;
+  for (char** e = ::environ; *e; e++) envCopy.emplace_back(*e);
   // the spawner thread does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
   Spawner::global().spawn(
-      loop, exec_, args_, raw_ ? -1 : pipefds[1],
+      loop, exec_, args_, std::move(envCopy), raw_ ? -1 : pipefds[1],
       [this, self, bus, readFd](pid_t pid, int err) {
         if (pid < 0) {
           LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(err));

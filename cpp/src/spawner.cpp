@@ -21,11 +21,12 @@ Spawner::Spawner() {
 }
 
 void Spawner::spawn(Loop& loop, std::string execPath,
-                    std::vector<std::string> args, int stdioFd, SpawnCb cb) {
+                    std::vector<std::string> args,
+                    std::vector<std::string> env, int stdioFd, SpawnCb cb) {
   {
     std::lock_guard<std::mutex> l(mu_);
     queue_.push_back(Request{&loop, std::move(execPath), std::move(args),
-                             stdioFd, std::move(cb)});
+                             std::move(env), stdioFd, std::move(cb)});
   }
   cv_.notify_one();
 }
@@ -63,10 +64,13 @@ void Spawner::threadMain() {
     argv.push_back(const_cast<char*>(req.execPath.c_str()));
     for (auto& a : req.args) argv.push_back(const_cast<char*>(a.c_str()));
     argv.push_back(nullptr);
+    std::vector<char*> envp;
+    for (auto& e : req.env) envp.push_back(const_cast<char*>(e.c_str()));
+    envp.push_back(nullptr);
 
     pid_t pid = -1;
     int rc = posix_spawnp(&pid, req.execPath.c_str(), &actions, &attr,
-                          argv.data(), environ);
+                          argv.data(), envp.data());
     posix_spawn_file_actions_destroy(&actions);
     posix_spawnattr_destroy(&attr);
     if (req.stdioFd >= 0) close(req.stdioFd);

@@ -263,3 +263,42 @@ def test_job_env_vars_exported(daemon_factory):
     m = re.search(r"PILOTPID=(\d+) MAINPID=(\d*)", log)
     assert m, log
     assert int(m.group(1)) == d.proc.pid
+
+
+def test_job_without_exec_no_crash(daemon_factory):
+    """A job with no exec is legal and doesn't break the lifecycle
+    (integration test_no_command)."""
+    d = daemon_factory(make_config(CONSUL, [
+        {"name": "noexec"},
+        {"name": "hello", "exec": "echo no-command-ok"},
+    ]))
+    d.start()
+    d.wait_for_socket()
+    time.sleep(0.3)
+    assert "no-command-ok" in d.log()
+    d.terminate()
+    assert d.wait(timeout=30) == 0
+
+
+def test_job_ip_env_exported(daemon_factory, mock_consul):
+    """CONTAINERPILOT_{JOB}_IP is exported for advertised jobs
+    (core/app.go:81-86, integration test_envvars)."""
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "jobs": [
+            {"name": "my-svc", "exec": "sleep 60", "port": 8123,
+             "interfaces": ["static:10.7.7.7"],
+             "health": {"exec": "true", "interval": 1, "ttl": 5}},
+            {"name": "ipdump",
+             "exec": ["sh", "-c", "echo SVCIP=$CONTAINERPILOT_MY_SVC_IP"]},
+        ],
+    })
+    d.start()
+    d.wait_for_socket()
+    deadline = time.time() + 10
+    while time.time() < deadline and "SVCIP=" not in d.log():
+        time.sleep(0.1)
+    assert "SVCIP=10.7.7.7" in d.log()
+    d.terminate()
+    assert d.wait(timeout=30) == 0

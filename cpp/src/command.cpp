@@ -139,6 +139,8 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
           LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(err));
           if (readFd >= 0) close(readFd);
           running_ = false;
+          pendingRun_ = false;
+          pendingSignal_ = 0;
           bus->publish(Event{EventCode::ExitFailed, name_});
           bus->publish(Event{EventCode::Error, strerror(err)});
           return;

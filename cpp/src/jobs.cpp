@@ -27,11 +27,6 @@ namespace {
 
 constexpr auto kTaskMinDuration = std::chrono::milliseconds(1);
 
-struct RawJob {
-  const Json* raw;
-  std::string name;
-};
-
 bool getRawBool(const Json* obj, const char* key, bool def) {
   if (!obj) return def;
   const Json* v = obj->find(key);

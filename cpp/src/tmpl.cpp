@@ -7,7 +7,6 @@
 #include <memory>
 #include <regex>
 #include <stdexcept>
-#include <variant>
 #include <vector>
 
 extern char** environ;

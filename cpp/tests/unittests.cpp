@@ -430,15 +430,6 @@ struct JobScenario {
                     [this] { loop.stop(); });
     loop.run();
   }
-
-  int countEvents(EventCode code, const std::string& source) {
-    // use the counter-free delivered history via debug ring? The ring
-    // only keeps 10; count from the events counter instead by scanning
-    // published history is not kept — so track via a probe subscriber.
-    (void)code;
-    (void)source;
-    return -1;
-  }
 };
 
 // probe subscriber recording every event it sees

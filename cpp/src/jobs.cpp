@@ -27,13 +27,18 @@ namespace {
 
 constexpr auto kTaskMinDuration = std::chrono::milliseconds(1);
 
-bool getRawBool(const Json* obj, const char* key, bool def) {
-  if (!obj) return def;
+// strict: a non-bool value is an error (mapstructure semantics)
+bool getRawBool(const Json* obj, const char* key, bool def, bool* out,
+                std::string* err) {
+  *out = def;
+  if (!obj) return true;
   const Json* v = obj->find(key);
-  if (!v) return def;
-  bool out = def;
-  decode::toBool(*v, &out);
-  return out;
+  if (!v) return true;
+  if (!decode::toBool(*v, out)) {
+    *err = std::string("cannot parse '") + key + "' as bool";
+    return false;
+  }
+  return true;
 }
 
 // parse when{} (jobs/config.go:178-246)
@@ -240,7 +245,12 @@ bool validateHealthCheck(const Json& raw, std::shared_ptr<JobConfig>& cfg,
   const Json* checkExec = health->find("exec");
   if (checkExec && !checkExec->isNull()) {
     std::string checkName = "check." + cfg->name;
-    bool raw_ = getRawBool(health->find("logging"), "raw", false);
+    bool raw_ = false;
+    std::string boolErr;
+    if (!getRawBool(health->find("logging"), "raw", false, &raw_, &boolErr)) {
+      *err = "job[" + cfg->name + "].health.logging: " + boolErr;
+      return false;
+    }
     std::string cmdErr;
     CommandPtr cmd =
         newCommand(*checkExec, checkTimeout, raw_, checkName, &cmdErr);
@@ -308,8 +318,13 @@ bool validateDiscovery(const Json& raw, ConsulBackend* disc,
         }
       }
     }
-    if (const Json* v = extras->find("enableTagOverride"))
-      decode::toBool(*v, &enableTagOverride);
+    if (const Json* v = extras->find("enableTagOverride")) {
+      if (!decode::toBool(*v, &enableTagOverride)) {
+        *err = "job configuration error: cannot parse "
+               "'enableTagOverride' as bool for job[" + cfg->name + "]";
+        return false;
+      }
+    }
   }
 
   auto svc = std::make_shared<ServiceDefinition>();
@@ -354,7 +369,12 @@ bool validateExec(const Json& raw, std::shared_ptr<JobConfig>& cfg,
   }
   const Json* exec = raw.find("exec");
   if (exec && !exec->isNull()) {
-    bool rawLog = getRawBool(raw.find("logging"), "raw", false);
+    bool rawLog = false;
+    std::string boolErr;
+    if (!getRawBool(raw.find("logging"), "raw", false, &rawLog, &boolErr)) {
+      *err = "job[" + cfg->name + "].logging: " + boolErr;
+      return false;
+    }
     std::string cmdErr;
     CommandPtr cmd =
         newCommand(*exec, cfg->execTimeout, rawLog, cfg->name, &cmdErr);

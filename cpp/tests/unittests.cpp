@@ -160,6 +160,18 @@ static void testTemplate() {
            std::string("y"));
   CHECK_EQ(renderTemplate("{{ if .TEST_MISSING }}y{{ else }}n{{ end }}"),
            std::string("n"));
+
+  // exact expectations from the reference's template_test.go:50-70
+  setenv("COUNT", "3", 1);
+  CHECK_EQ(renderTemplate("{{ loop 2 5 }}"), std::string("[2 3 4]"));
+  CHECK_EQ(renderTemplate("{{ loop 10 1 }}"),
+           std::string("[10 9 8 7 6 5 4 3 2]"));
+  CHECK_EQ(renderTemplate("{{ loop 5 }}"), std::string("[0 1 2 3 4]"));
+  CHECK_EQ(renderTemplate("{{ loop .COUNT }}"), std::string("[0 1 2]"));
+  CHECK_EQ(renderTemplate("{{ loop 1 .COUNT }}"), std::string("[1 2]"));
+  CHECK_EQ(renderTemplate("{{ range $i := loop 2 5 -}}i={{$i}},{{ end }}"),
+           std::string("i=2,i=3,i=4,"));
+  unsetenv("COUNT");
 }
 
 static void testEvents() {

@@ -1,0 +1,109 @@
+"""Config fidelity against the reference's own test fixture corpus.
+
+The fixtures are read from the read-only reference checkout at test
+time (never copied into this repo); when the checkout is absent (e.g.
+on the GPU box) these tests skip. Expected outcomes mirror the
+reference's own unit tests (jobs/config_test.go, watches/config_test.go,
+telemetry/telemetry_config_test.go).
+"""
+
+import json
+import os
+
+import pytest
+
+from containerpilot_amd import native
+
+REF = "/root/reference"
+
+pytestmark = pytest.mark.skipif(
+    not os.path.isdir(REF), reason="reference checkout not available")
+
+
+def load(rel):
+    with open(os.path.join(REF, rel)) as f:
+        return f.read()
+
+
+def validate_jobs_array(raw_array_text):
+    """Wrap a jobs-array fixture in a minimal full config and validate."""
+    arr = json.loads(native.json5_to_json(raw_array_text))
+    cfg = json.dumps({"consul": "localhost:8500", "jobs": arr})
+    return native.validate_config(cfg)
+
+
+def test_job_smoke_fixture_valid():
+    # jobs/config_test.go:TestJobConfigSmokeTest expects this to parse
+    err = validate_jobs_array(load("jobs/testdata/TestJobConfigSmokeTest.json5"))
+    assert err is None, err
+
+
+@pytest.mark.parametrize("fixture", [
+    "TestJobConfigPeriodicTask",
+    "TestJobConfigServiceWithPreStart",
+    "TestJobConfigServiceWithStopping",
+    "TestJobConfigServiceWithArrayExec",
+    "TestJobConfigServiceWithInitialStatus",
+    "TestJobConfigConsulExtras",
+    "TestJobConfigHealthTimeout",
+    "TestJobConfigServiceNonAdvertised",
+])
+def test_job_fixtures_valid(fixture):
+    err = validate_jobs_array(load("jobs/testdata/%s.json5" % fixture))
+    assert err is None, "%s: %s" % (fixture, err)
+
+
+@pytest.mark.parametrize("fixture,why", [
+    ("TestErrJobConfigConsulDeregisterCriticalServiceAfter",
+     "bad duration 'nope'"),
+    ("TestErrJobConfigConsulEnableTagOverride",
+     "enableTagOverride must be a bool"),
+])
+def test_job_fixtures_invalid(fixture, why):
+    err = validate_jobs_array(load("jobs/testdata/%s.json5" % fixture))
+    assert err is not None, "%s should be rejected (%s)" % (fixture, why)
+
+
+def test_watches_fixture_valid():
+    arr = json.loads(native.json5_to_json(
+        load("watches/testdata/TestWatchesParse.json5")))
+    cfg = json.dumps({"consul": "localhost:8500", "watches": arr})
+    assert native.validate_config(cfg) is None
+
+
+def test_telemetry_fixture_valid():
+    tel = json.loads(native.json5_to_json(
+        load("telemetry/testdata/TestTelemetryConfigParse.json5")))
+    cfg = json.dumps({"consul": "localhost:8500", "telemetry": tel})
+    assert native.validate_config(cfg) is None
+
+
+def test_app_testcfg_shapes():
+    """The config shapes core/app_test.go feeds NewApp."""
+    # valid minimal reload config (core/app_test.go:128)
+    assert native.validate_config('{ "consul": "newconsul:8500" }') is None
+    # missing job name (core/app_test.go:25-30)
+    err = native.validate_config(
+        '{"consul": "consul:8500", "jobs": ['
+        '{"name": "", "port": 8080, '
+        '"health": {"interval": 30, "ttl": 19}}]}')
+    assert err and "'name' must not be blank" in err
+    # missing health interval (core/app_test.go:33-38)
+    err = native.validate_config(
+        '{"consul": "consul:8500", "jobs": ['
+        '{"name": "name", "port": 8080, "health": {"ttl": 19}}]}')
+    assert err and "health.interval must be > 0" in err
+    # missing ttl
+    err = native.validate_config(
+        '{"consul": "consul:8500", "jobs": ['
+        '{"name": "name", "port": 8080, "health": {"interval": 19}}]}')
+    assert err and "health.ttl must be > 0" in err
+    # missing watch name (core/app_test.go:51-55)
+    err = native.validate_config(
+        '{"consul": "consul:8500", "watches": ['
+        '{"name": "", "interval": 30}]}')
+    assert err and "'name' must not be blank" in err
+    # missing watch interval
+    err = native.validate_config(
+        '{"consul": "consul:8500", "watches": [{"name": "name"}]}')
+    assert err and "interval must be > 0" in err

@@ -21,7 +21,7 @@ from urllib.parse import urlparse, parse_qs
 
 
 class MockConsul:
-    def __init__(self, port=0):
+    def __init__(self, port=0, certfile=None, keyfile=None):
         self.lock = threading.Lock()
         self.services = {}      # id -> registration payload
         self.ttl_updates = []   # (checkID, payload)
@@ -105,6 +105,12 @@ class MockConsul:
             do_POST = do_PUT
 
         self.server = ThreadingHTTPServer(("127.0.0.1", port), Handler)
+        if certfile:
+            import ssl
+            ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            ctx.load_cert_chain(certfile, keyfile)
+            self.server.socket = ctx.wrap_socket(self.server.socket,
+                                                 server_side=True)
         self.port = self.server.server_address[1]
         self.thread = threading.Thread(target=self.server.serve_forever,
                                        daemon=True)

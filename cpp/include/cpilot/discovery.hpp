@@ -15,6 +15,7 @@
 #include <thread>
 #include <vector>
 
+#include "cpilot/http.hpp"
 #include "cpilot/json.hpp"
 #include "cpilot/loop.hpp"
 #include "cpilot/metrics.hpp"
@@ -77,6 +78,7 @@ class ConsulBackend {
   std::string address_;  // host:port
   std::string scheme_ = "http";
   std::string token_;
+  http::TlsOptions tls_;
 
   Loop* loop_ = nullptr;
   std::vector<std::thread> workers_;

@@ -69,12 +69,25 @@ struct ClientResult {
   std::string error;
 };
 
+// TLS options for https targets (OpenSSL; reference parity with the
+// consul api.TLSConfig fields, discovery/config.go:29-61)
+struct TlsOptions {
+  bool enabled = false;
+  std::string caFile;
+  std::string caPath;
+  std::string certFile;
+  std::string keyFile;
+  std::string serverName;          // SNI + hostname verification override
+  bool insecureSkipVerify = false;
+};
+
 // target: "unix:<path>" or "host:port"
 ClientResult request(const std::string& target, const std::string& method,
                      const std::string& path, const std::string& body,
                      const std::string& contentType = "application/json",
                      const std::map<std::string, std::string>& headers = {},
-                     int timeoutMs = 10000);
+                     int timeoutMs = 10000,
+                     const TlsOptions* tls = nullptr);
 
 }  // namespace http
 }  // namespace cpilot

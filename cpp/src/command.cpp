@@ -193,15 +193,17 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
           });
         }
 
+        // a term/kill that arrived while the spawn was in flight:
+        // deliver before watchChild, which may synchronously run
+        // onExit (and even a queued respawn) for an already-reaped pid
+        if (pendingSignal_ != 0) {
+          ::kill(-pid, pendingSignal_);
+          pendingSignal_ = 0;
+        }
+
         loop_->watchChild(pid, [this, self, bus](int status) {
           onExit(*loop_, bus, status);
         });
-
-        // a term/kill that arrived while the spawn was in flight
-        if (pendingSignal_ != 0 && pid_ > 0) {
-          ::kill(-pid_, pendingSignal_);
-          pendingSignal_ = 0;
-        }
       });
 }
 

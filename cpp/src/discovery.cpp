@@ -3,6 +3,7 @@
 #include <algorithm>
 #include <cstdlib>
 
+#include "cpilot/decode.hpp"
 #include "cpilot/http.hpp"
 #include "cpilot/log.hpp"
 
@@ -59,9 +60,30 @@ std::unique_ptr<ConsulBackend> ConsulBackend::create(const Json* raw,
       else if (kv.first == "token" && kv.second.isString())
         backend->token_ = kv.second.str();
       else if (kv.first == "tls") {
-        // TLS options are accepted for config compatibility; https
-        // transport is not implemented in this build and fails at
-        // request time with a clear error.
+        // optional TLS settings (discovery/config.go:18-25)
+        if (kv.second.isObject()) {
+          for (auto& tkv : kv.second.object()) {
+            if (tkv.first == "cafile" && tkv.second.isString())
+              backend->tls_.caFile = tkv.second.str();
+            else if (tkv.first == "capath" && tkv.second.isString())
+              backend->tls_.caPath = tkv.second.str();
+            else if (tkv.first == "clientcert" && tkv.second.isString())
+              backend->tls_.certFile = tkv.second.str();
+            else if (tkv.first == "clientkey" && tkv.second.isString())
+              backend->tls_.keyFile = tkv.second.str();
+            else if (tkv.first == "servername" && tkv.second.isString())
+              backend->tls_.serverName = tkv.second.str();
+            else if (tkv.first == "verify") {
+              bool verify = false;
+              if (decode::toBool(tkv.second, &verify))
+                backend->tls_.insecureSkipVerify = !verify;
+            } else {
+              *err = "consul configuration error: invalid tls key: " +
+                     tkv.first;
+              return nullptr;
+            }
+          }
+        }
       } else {
         *err = "consul configuration error: invalid key: " + kv.first;
         return nullptr;
@@ -79,6 +101,25 @@ std::unique_ptr<ConsulBackend> ConsulBackend::create(const Json* raw,
     if (addr[0] && raw->isString() && raw->str().empty())
       parseRawURI(addr, &backend->address_, &backend->scheme_);
   }
+  // env overrides (discovery/config.go:29-51)
+  if (const char* v = getenv("CONSUL_CACERT"))
+    if (v[0]) backend->tls_.caFile = v;
+  if (const char* v = getenv("CONSUL_CAPATH"))
+    if (v[0]) backend->tls_.caPath = v;
+  if (const char* v = getenv("CONSUL_CLIENT_CERT"))
+    if (v[0]) backend->tls_.certFile = v;
+  if (const char* v = getenv("CONSUL_CLIENT_KEY"))
+    if (v[0]) backend->tls_.keyFile = v;
+  if (const char* v = getenv("CONSUL_TLS_SERVER_NAME"))
+    if (v[0]) backend->tls_.serverName = v;
+  if (const char* v = getenv("CONSUL_HTTP_SSL_VERIFY")) {
+    std::string s2(v);
+    for (auto& ch : s2) ch = tolower(ch);
+    if (s2 == "1" || s2 == "true") backend->tls_.insecureSkipVerify = false;
+    else if (s2 == "0" || s2 == "false")
+      backend->tls_.insecureSkipVerify = true;
+  }
+  backend->tls_.enabled = (backend->scheme_ == "https");
   backend->watchGauge_ = prom::Registry::global().registerFamily(
       "containerpilot_watch_instances",
       "gauge of instances found for each ContainerPilot watch, partitioned "
@@ -160,18 +201,13 @@ void ConsulBackend::serviceRegister(
 
   std::string target = address_;
   std::string token = token_;
-  std::string scheme = scheme_;
+  http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, scheme, body, cb, loop] {
+  enqueue([target, token, tls, body, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
-    http::ClientResult res;
-    if (scheme == "https") {
-      res.error = "https transport not supported";
-    } else {
-      res = http::request(target, "PUT", "/v1/agent/service/register", body,
-                          "application/json", headers);
-    }
+    auto res = http::request(target, "PUT", "/v1/agent/service/register",
+                             body, "application/json", headers, 10000, &tls);
     bool ok = res.ok && res.status == 200;
     std::string err = res.ok ? ("status " + std::to_string(res.status) + ": " +
                                 res.body)
@@ -196,12 +232,13 @@ void ConsulBackend::updateTTL(const std::string& checkID,
 
   std::string target = address_;
   std::string token = token_;
+  http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, path, body, cb, loop] {
+  enqueue([target, token, tls, path, body, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "PUT", path, body, "application/json",
-                             headers);
+                             headers, 10000, &tls);
     bool ok = res.ok && res.status == 200;
     std::string err =
         res.ok ? ("status " + std::to_string(res.status) + ": " + res.body)
@@ -214,12 +251,13 @@ void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
   std::string path = "/v1/agent/service/deregister/" + urlEncode(id);
   std::string target = address_;
   std::string token = token_;
+  http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, path, cb, loop] {
+  enqueue([target, token, tls, path, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "PUT", path, "", "application/json",
-                             headers);
+                             headers, 10000, &tls);
     bool ok = res.ok && res.status == 200;
     std::string err =
         res.ok ? ("status " + std::to_string(res.status)) : res.error;
@@ -235,12 +273,13 @@ void ConsulBackend::healthService(const std::string& name,
   if (!dc.empty()) path += "&dc=" + urlEncode(dc);
   std::string target = address_;
   std::string token = token_;
+  http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, path, cb, loop] {
+  enqueue([target, token, tls, path, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "GET", path, "", "application/json",
-                             headers);
+                             headers, 10000, &tls);
     bool ok = res.ok && res.status == 200;
     std::vector<ServiceEntry> entries;
     if (ok) {

@@ -8,6 +8,8 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <poll.h>
+#include <openssl/err.h>
+#include <openssl/ssl.h>
 #include <sys/socket.h>
 #include <sys/un.h>
 #include <unistd.h>
@@ -297,16 +299,94 @@ int connectTarget(const std::string& target, int timeoutMs, std::string* err) {
 
 }  // namespace
 
+namespace {
+
+// thin transport abstraction so the request logic is shared between
+// plain sockets and TLS sessions
+struct Transport {
+  int fd = -1;
+  SSL_CTX* ctx = nullptr;
+  SSL* ssl = nullptr;
+
+  ~Transport() {
+    if (ssl) {
+      SSL_shutdown(ssl);
+      SSL_free(ssl);
+    }
+    if (ctx) SSL_CTX_free(ctx);
+    if (fd >= 0) close(fd);
+  }
+
+  bool startTls(const std::string& host, const TlsOptions& tls,
+                std::string* err) {
+    ctx = SSL_CTX_new(TLS_client_method());
+    if (!ctx) {
+      *err = "SSL_CTX_new failed";
+      return false;
+    }
+    if (!tls.caFile.empty() || !tls.caPath.empty()) {
+      if (SSL_CTX_load_verify_locations(
+              ctx, tls.caFile.empty() ? nullptr : tls.caFile.c_str(),
+              tls.caPath.empty() ? nullptr : tls.caPath.c_str()) != 1) {
+        *err = "failed to load CA certificates";
+        return false;
+      }
+    } else {
+      SSL_CTX_set_default_verify_paths(ctx);
+    }
+    if (!tls.certFile.empty() &&
+        (SSL_CTX_use_certificate_chain_file(ctx, tls.certFile.c_str()) != 1 ||
+         SSL_CTX_use_PrivateKey_file(ctx, tls.keyFile.c_str(),
+                                     SSL_FILETYPE_PEM) != 1)) {
+      *err = "failed to load client certificate/key";
+      return false;
+    }
+    SSL_CTX_set_verify(
+        ctx, tls.insecureSkipVerify ? SSL_VERIFY_NONE : SSL_VERIFY_PEER,
+        nullptr);
+    ssl = SSL_new(ctx);
+    SSL_set_fd(ssl, fd);
+    std::string sniHost = tls.serverName.empty() ? host : tls.serverName;
+    // strip :port for SNI/verification
+    size_t colon = sniHost.rfind(':');
+    if (colon != std::string::npos) sniHost = sniHost.substr(0, colon);
+    SSL_set_tlsext_host_name(ssl, sniHost.c_str());
+    if (!tls.insecureSkipVerify) SSL_set1_host(ssl, sniHost.c_str());
+    if (SSL_connect(ssl) != 1) {
+      unsigned long e = ERR_get_error();
+      char ebuf[256];
+      ERR_error_string_n(e, ebuf, sizeof(ebuf));
+      *err = std::string("TLS handshake failed: ") + ebuf;
+      return false;
+    }
+    return true;
+  }
+
+  ssize_t send(const char* data, size_t len) {
+    return ssl ? SSL_write(ssl, data, (int)len) : write(fd, data, len);
+  }
+  ssize_t recv(char* data, size_t len) {
+    return ssl ? SSL_read(ssl, data, (int)len) : read(fd, data, len);
+  }
+};
+
+}  // namespace
+
 ClientResult request(const std::string& target, const std::string& method,
                      const std::string& path, const std::string& body,
                      const std::string& contentType,
                      const std::map<std::string, std::string>& headers,
-                     int timeoutMs) {
+                     int timeoutMs, const TlsOptions* tls) {
   ClientResult result;
-  int fd = connectTarget(target, timeoutMs, &result.error);
-  if (fd < 0) return result;
+  Transport t;
+  t.fd = connectTarget(target, timeoutMs, &result.error);
+  if (t.fd < 0) return result;
 
   std::string host = target.rfind("unix:", 0) == 0 ? "localhost" : target;
+  if (tls && tls->enabled) {
+    if (!t.startTls(host, *tls, &result.error)) return result;
+  }
+
   std::string req = method + " " + path + " HTTP/1.1\r\n";
   req += "Host: " + host + "\r\n";
   req += "Connection: close\r\n";
@@ -320,11 +400,10 @@ ClientResult request(const std::string& target, const std::string& method,
 
   size_t off = 0;
   while (off < req.size()) {
-    ssize_t n = write(fd, req.data() + off, req.size() - off);
+    ssize_t n = t.send(req.data() + off, req.size() - off);
     if (n <= 0) {
-      if (n < 0 && errno == EINTR) continue;
+      if (n < 0 && errno == EINTR && !t.ssl) continue;
       result.error = "write failed";
-      close(fd);
       return result;
     }
     off += n;
@@ -333,16 +412,15 @@ ClientResult request(const std::string& target, const std::string& method,
   std::string resp;
   char buf[8192];
   while (true) {
-    ssize_t n = read(fd, buf, sizeof(buf));
+    ssize_t n = t.recv(buf, sizeof(buf));
     if (n > 0) {
       resp.append(buf, n);
-    } else if (n < 0 && errno == EINTR) {
+    } else if (n < 0 && errno == EINTR && !t.ssl) {
       continue;
     } else {
       break;
     }
   }
-  close(fd);
 
   size_t headerEnd = resp.find("\r\n\r\n");
   if (headerEnd == std::string::npos || resp.compare(0, 5, "HTTP/") != 0) {

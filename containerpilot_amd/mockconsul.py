@@ -48,7 +48,7 @@ class MockConsul:
             def do_GET(self):
                 parsed = urlparse(self.path)
                 with outer.lock:
-                    outer.requests.append(("GET", parsed.path))
+                    outer.requests.append(("GET", self.path))
                     tok = self.headers.get("X-Consul-Token")
                     if tok:
                         outer.tokens.append(tok)

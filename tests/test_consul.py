@@ -166,3 +166,23 @@ def test_watch_no_change_no_events(daemon_factory, mock_consul):
     assert d.log().count("{StatusChanged watch.stable}") == 1
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_watch_tag_and_dc_in_query(daemon_factory, mock_consul):
+    """watch tag/dc flow into the health query string
+    (discovery/consul.go:87-89: Health().Service(name, tag, passing,
+    {Datacenter: dc}))."""
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "jobs": [{"name": "main-app", "exec": "sleep 60"}],
+        "watches": [{"name": "tagged", "interval": 1, "tag": "prod",
+                     "dc": "us-east-1"}],
+    }).start()
+    d.wait_for_socket()
+    assert wait_until(lambda: any(
+        "/v1/health/service/tagged" in path and "passing=1" in path and
+        "tag=prod" in path and "dc=us-east-1" in path
+        for _, path in mock_consul.requests)), mock_consul.requests[-5:]
+    d.terminate()
+    assert d.wait(timeout=30) == 0

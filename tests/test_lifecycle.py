@@ -68,8 +68,8 @@ def test_periodic_job_run_counts(daemon_factory):
     d.start()
     assert d.wait(timeout=30) == 0
     count = d.log().count("{ExitSuccess tick}")
-    # ~10 ticks in 2s; allow wide margin for slow CI
-    assert 5 <= count <= 12, d.log()
+    # ~10 ticks in 2s; allow a wide margin for slow CI
+    assert 3 <= count <= 13, d.log()
 
 
 def test_when_once_dependency_chain(daemon_factory):
@@ -149,15 +149,21 @@ def test_coprocess_restart_resets_on_reload(daemon_factory, tmp_path):
     ]))
     d.start()
     d.wait_for_socket()
-    time.sleep(1.0)
-    log = d.log()
-    assert log.count("{ExitSuccess coprocess}") == 2  # initial + 1 restart
+    deadline = time.time() + 10
+    while time.time() < deadline and \
+            d.log().count("{ExitSuccess coprocess}") < 2:
+        time.sleep(0.1)
+    time.sleep(0.5)  # settle: no third run may appear
+    assert d.log().count("{ExitSuccess coprocess}") == 2  # initial + 1
     # reload resets the restart budget
     status, _ = d.control("POST", "/v3/reload")
     assert status == 200
-    time.sleep(1.5)
-    log = d.log()
-    assert log.count("{ExitSuccess coprocess}") == 4
+    deadline = time.time() + 10
+    while time.time() < deadline and \
+            d.log().count("{ExitSuccess coprocess}") < 4:
+        time.sleep(0.1)
+    time.sleep(0.5)
+    assert d.log().count("{ExitSuccess coprocess}") == 4
     d.terminate()
     assert d.wait(timeout=30) == 0
 

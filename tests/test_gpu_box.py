@@ -50,10 +50,13 @@ def test_daemon_lifecycle_on_gpu_box():
 @pytest.mark.gpu
 def test_stress_meets_baseline_on_gpu_box():
     """BASELINE stress config on the box: >=10k events/sec through the
-    bus and <1ms p99 dispatch latency."""
+    bus and <1ms p99 dispatch latency, measured over a post-warmup
+    window exactly like bench.py (whole-run stats include the startup
+    registration burst, which is excluded from the steady-state
+    targets)."""
     sys.path.insert(0, os.path.dirname(os.path.dirname(
         os.path.abspath(__file__))))
-    from bench import stress_config, free_port
+    from bench import stress_config, free_port, scrape, histogram_p99
     from containerpilot_amd.mockconsul import MockConsul
 
     mc = MockConsul().start()
@@ -65,16 +68,22 @@ def test_stress_meets_baseline_on_gpu_box():
     port = free_port()
     cfg = stress_config(mc.address, port, 100, 50, 100,
                         os.path.join(wd, "cp.socket"))
-    d = harness.Daemon(config_dict=cfg, workdir=wd,
-                       extra_args=["-bench-seconds", "15"])
+    d = harness.Daemon(config_dict=cfg, workdir=wd)
     try:
         d.start()
         d.wait_for_socket(timeout=30)
+        time.sleep(5)  # warmup: exclude the startup/registration burst
+        t0 = time.monotonic()
+        s0 = scrape(port)
+        time.sleep(10)
+        s1 = scrape(port)
+        elapsed = time.monotonic() - t0
+        delivered_per_sec = (s1["delivered"] - s0["delivered"]) / elapsed
+        p99 = histogram_p99(s0["buckets"], s1["buckets"])
+        assert delivered_per_sec >= 10000, (delivered_per_sec, s0, s1)
+        assert p99 is not None and p99 * 1e3 < 1.0, p99
+        d.terminate()
         assert d.wait(timeout=60) == 0
-        st = d.stats()
-        delivered_per_sec = st["events_delivered"] / st["wall_seconds"]
-        assert delivered_per_sec >= 10000, st
-        assert st["dispatch_p99_us"] < 1000, st
     finally:
         d.cleanup()
         mc.stop()

@@ -122,12 +122,17 @@ class Registry {
  public:
   static Registry& global();
 
-  // Registers (or replaces, matching the reference's unregister-then-register
-  // reload behavior, metrics_config.go:83-85) a family and returns it.
+  // Registers a family and returns it. keepExisting=true returns the
+  // already-registered family of that name (internal collectors persist
+  // across config reloads, like the reference's package-level
+  // prometheus.MustRegister in init()); keepExisting=false replaces it
+  // (user metrics unregister-then-register on reload and reset,
+  // metrics_config.go:83-85).
   std::shared_ptr<Family> registerFamily(const std::string& name,
                                          const std::string& help,
                                          MetricType type,
-                                         std::vector<std::string> labelNames = {});
+                                         std::vector<std::string> labelNames = {},
+                                         bool keepExisting = true);
   std::string expose() const;
 
  private:

@@ -183,12 +183,16 @@ void Server::connReadable(std::shared_ptr<Conn> c) {
   out += "Connection: close\r\n\r\n";
   out += resp.body;
   size_t off = 0;
+  int stalls = 0;
   while (off < out.size()) {
     ssize_t n = write(c->fd, out.data() + off, out.size() - off);
     if (n > 0) {
       off += n;
+      stalls = 0;
     } else if (n < 0 && (errno == EAGAIN || errno == EINTR)) {
-      // short blocking wait; responses are small
+      // short blocking wait; responses are small. A peer that stops
+      // reading gets dropped after ~3s rather than wedging the server.
+      if (++stalls > 3) break;
       struct pollfd pfd{c->fd, POLLOUT, 0};
       poll(&pfd, 1, 1000);
     } else {

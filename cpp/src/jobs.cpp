@@ -50,6 +50,10 @@ bool validateWhen(const Json* when, std::shared_ptr<JobConfig>& cfg,
     cfg->whenStartsLimit = 1;
     return true;
   }
+  if (!when->isObject()) {
+    *err = "job[" + cfg->name + "].when must be an object";
+    return false;
+  }
   if (!decode::checkKeys(*when,
                          {"interval", "source", "once", "each", "timeout"},
                          err)) {
@@ -192,6 +196,10 @@ bool validateHealthCheck(const Json& raw, std::shared_ptr<JobConfig>& cfg,
     return false;
   }
   if (!health || health->isNull()) return true;
+  if (!health->isObject()) {
+    *err = "job[" + cfg->name + "].health must be an object";
+    return false;
+  }
   if (!decode::checkKeys(*health,
                          {"exec", "timeout", "interval", "ttl", "logging"},
                          err)) {
@@ -300,6 +308,10 @@ bool validateDiscovery(const Json& raw, ConsulBackend* disc,
   std::string deregAfter;
   const Json* extras = raw.find("consul");
   if (extras && !extras->isNull()) {
+    if (!extras->isObject()) {
+      *err = "job[" + cfg->name + "].consul must be an object";
+      return false;
+    }
     if (!decode::checkKeys(
             *extras, {"enableTagOverride", "deregisterCriticalServiceAfter"},
             err)) {
@@ -423,6 +435,10 @@ bool validateJobConfig(const Json& raw, ConsulBackend* disc,
     }
   }
   if (const Json* logging = raw.find("logging")) {
+    if (!logging->isNull() && !logging->isObject()) {
+      *err = "job[" + cfg->name + "].logging must be an object";
+      return false;
+    }
     if (logging->isObject() &&
         !decode::checkKeys(*logging, {"raw"}, err)) {
       *err = "job configuration error: " + *err;

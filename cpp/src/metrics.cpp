@@ -117,10 +117,11 @@ Registry& Registry::global() {
 
 std::shared_ptr<Family> Registry::registerFamily(
     const std::string& name, const std::string& help, MetricType type,
-    std::vector<std::string> labelNames) {
+    std::vector<std::string> labelNames, bool keepExisting) {
   std::lock_guard<std::mutex> l(mu_);
   for (auto it = families_.begin(); it != families_.end(); ++it) {
     if ((*it)->name() == name) {
+      if (keepExisting && (*it)->type() == type) return *it;
       families_.erase(it);
       break;
     }

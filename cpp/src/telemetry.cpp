@@ -65,7 +65,7 @@ bool newMetricConfigs(const Json& raw,
     // unregister-then-register so reloads survive
     // (telemetry/metrics_config.go:83-85)
     cfg->collector = prom::Registry::global().registerFamily(
-        fq, cfg->help, cfg->metricType);
+        fq, cfg->help, cfg->metricType, {}, /*keepExisting=*/false);
     out->push_back(cfg);
   }
   return true;

@@ -29,6 +29,7 @@ class MockConsul:
         self.requests = []      # (method, path)
         self.health = {}        # service name -> list of dicts(ID,Address,Port)
         self.tokens = []        # X-Consul-Token header values seen
+        self.fail_mode = False  # when True: every endpoint returns 500
 
         outer = self
 
@@ -47,6 +48,8 @@ class MockConsul:
 
             def do_GET(self):
                 parsed = urlparse(self.path)
+                if outer.fail_mode:
+                    return self._respond(500, b"boom")
                 with outer.lock:
                     outer.requests.append(("GET", self.path))
                     tok = self.headers.get("X-Consul-Token")
@@ -77,6 +80,8 @@ class MockConsul:
 
             def do_PUT(self):
                 parsed = urlparse(self.path)
+                if outer.fail_mode:
+                    return self._respond(500, b"boom")
                 length = int(self.headers.get("Content-Length", 0))
                 raw = self.rfile.read(length) if length else b""
                 payload = json.loads(raw) if raw else {}

@@ -10,6 +10,7 @@
 #pragma once
 
 #include <deque>
+#include <random>
 #include <functional>
 #include <memory>
 #include <string>
@@ -119,7 +120,9 @@ class Bus : public std::enable_shared_from_this<Bus> {
   std::shared_ptr<prom::Family> dispatchHist_;
   std::shared_ptr<prom::Family> deliveriesCounter_;
   std::vector<double> latencyWindow_;
-  size_t latencyCap_ = 262144;
+  size_t latencyCap_ = 65536;
+  uint64_t latencySeen_ = 0;
+  std::minstd_rand latencyRng_{12345};
 };
 
 }  // namespace cpilot

@@ -1,5 +1,6 @@
 #include "cpilot/app.hpp"
 
+#include <malloc.h>
 #include <signal.h>
 #include <sys/epoll.h>
 #include <sys/signalfd.h>
@@ -198,8 +199,21 @@ void App::teardownGeneration() {
 }
 
 int App::run() {
+  // long-running daemon hygiene: cap glibc arenas (worker threads
+  // otherwise each grow their own) and periodically return freed pages
+  // to the OS so RSS tracks live data instead of allocator high-water
+  mallopt(M_ARENA_MAX, 2);
+  loop_.addInterval(std::chrono::seconds(60), [] { malloc_trim(0); });
+
   setupSignals();
   startTime_ = Clock::now();
+  if (getenv("CPILOT_MEMDEBUG")) {
+    loop_.addInterval(std::chrono::seconds(10), [] {
+      struct mallinfo2 mi = mallinfo2();
+      LOG_WARN("memdebug: heap_alloc=%zu KB arena=%zu KB free=%zu KB",
+               mi.uordblks / 1024, mi.arena / 1024, mi.fordblks / 1024);
+    });
+  }
   while (true) {
     startGeneration();
     loop_.run();

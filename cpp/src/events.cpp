@@ -118,7 +118,16 @@ void Bus::drain() {
     double latency =
         std::chrono::duration<double>(Clock::now() - publishedAt).count();
     dispatchHist_->observe(latency);
-    if (latencyWindow_.size() < latencyCap_) latencyWindow_.push_back(latency);
+    // reservoir sample: bounded memory with uniform coverage of the
+    // whole run (a plain prefix window stopped representing steady
+    // state and grew ~16 KB/s until its cap)
+    latencySeen_++;
+    if (latencyWindow_.size() < latencyCap_) {
+      latencyWindow_.push_back(latency);
+    } else {
+      uint64_t slot = latencyRng_() % latencySeen_;
+      if (slot < latencyCap_) latencyWindow_[slot] = latency;
+    }
     // handlers may (un)subscribe during delivery: unsubscribes tombstone
     // their slot (checked per delivery), new subscribes append past the
     // bound captured here so they don't see this event

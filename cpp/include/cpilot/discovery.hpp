@@ -10,6 +10,7 @@
 #include <functional>
 #include <map>
 #include <memory>
+#include <set>
 #include <mutex>
 #include <string>
 #include <thread>
@@ -73,7 +74,12 @@ class ConsulBackend {
  private:
   ConsulBackend() = default;
   void workerMain();
-  void enqueue(std::function<void()> task);
+  // key != "": coalesce — if a task with the same key is already
+  // queued, the new one is rejected (idempotent heartbeats/registrations
+  // must not pile up behind a slow agent). The queue is also hard-capped.
+  // Returns false when rejected; callers must complete their callback
+  // path themselves (in-flight guards would otherwise stick).
+  bool enqueue(const std::string& key, std::function<void()> task);
 
   std::string address_;  // host:port
   std::string scheme_ = "http";
@@ -82,7 +88,9 @@ class ConsulBackend {
 
   Loop* loop_ = nullptr;
   std::vector<std::thread> workers_;
-  std::deque<std::function<void()>> tasks_;
+  std::deque<std::pair<std::string, std::function<void()>>> tasks_;
+  std::set<std::string> queuedKeys_;
+  uint64_t dropped_ = 0;
   std::mutex mu_;
   std::condition_variable cv_;
   bool stopping_ = false;

@@ -155,20 +155,35 @@ void ConsulBackend::workerMain() {
       std::unique_lock<std::mutex> l(mu_);
       cv_.wait(l, [this] { return stopping_ || !tasks_.empty(); });
       if (stopping_ && tasks_.empty()) return;
-      task = std::move(tasks_.front());
+      auto& front = tasks_.front();
+      if (!front.first.empty()) queuedKeys_.erase(front.first);
+      task = std::move(front.second);
       tasks_.pop_front();
     }
     task();
   }
 }
 
-void ConsulBackend::enqueue(std::function<void()> task) {
+bool ConsulBackend::enqueue(const std::string& key,
+                            std::function<void()> task) {
   {
     std::lock_guard<std::mutex> l(mu_);
-    if (stopping_) return;
-    tasks_.push_back(std::move(task));
+    if (stopping_) return false;
+    // coalesce: a second heartbeat/registration for the same target is
+    // redundant while one is still queued behind a slow agent
+    if (!key.empty() && !queuedKeys_.insert(key).second) return false;
+    // hard cap: a dead-slow agent must not grow the daemon unboundedly
+    if (tasks_.size() >= 10000) {
+      if (++dropped_ % 1000 == 1)
+        LOG_WARN("consul task queue full, dropping requests (%llu dropped)",
+                 (unsigned long long)dropped_);
+      if (!key.empty()) queuedKeys_.erase(key);
+      return false;
+    }
+    tasks_.emplace_back(key, std::move(task));
   }
   cv_.notify_one();
+  return true;
 }
 
 void ConsulBackend::serviceRegister(
@@ -203,7 +218,7 @@ void ConsulBackend::serviceRegister(
   std::string token = token_;
   http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, tls, body, cb, loop] {
+  bool accepted = enqueue("reg:" + id, [target, token, tls, body, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "PUT", "/v1/agent/service/register",
@@ -214,6 +229,8 @@ void ConsulBackend::serviceRegister(
                              : res.error;
     loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
   });
+  if (!accepted)
+    loop_->post([cb] { cb(false, "request coalesced or queue full"); });
 }
 
 void ConsulBackend::updateTTL(const std::string& checkID,
@@ -234,7 +251,8 @@ void ConsulBackend::updateTTL(const std::string& checkID,
   std::string token = token_;
   http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, tls, path, body, cb, loop] {
+  bool accepted =
+      enqueue("ttl:" + checkID, [target, token, tls, path, body, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "PUT", path, body, "application/json",
@@ -245,6 +263,10 @@ void ConsulBackend::updateTTL(const std::string& checkID,
                : res.error;
     loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
   });
+  if (!accepted) {
+    // a heartbeat for this check is already queued; nothing to report
+    (void)cb;
+  }
 }
 
 void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
@@ -253,7 +275,7 @@ void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
   std::string token = token_;
   http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, tls, path, cb, loop] {
+  bool accepted = enqueue("", [target, token, tls, path, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "PUT", path, "", "application/json",
@@ -263,6 +285,7 @@ void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
         res.ok ? ("status " + std::to_string(res.status)) : res.error;
     loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
   });
+  if (!accepted) loop_->post([cb] { cb(false, "queue full"); });
 }
 
 void ConsulBackend::healthService(const std::string& name,
@@ -275,7 +298,8 @@ void ConsulBackend::healthService(const std::string& name,
   std::string token = token_;
   http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  enqueue([target, token, tls, path, cb, loop] {
+  bool accepted = enqueue("health:" + name + "|" + tag + "|" + dc,
+                          [target, token, tls, path, cb, loop] {
     std::map<std::string, std::string> headers;
     if (!token.empty()) headers["X-Consul-Token"] = token;
     auto res = http::request(target, "GET", path, "", "application/json",
@@ -307,6 +331,9 @@ void ConsulBackend::healthService(const std::string& name,
       cb(ok, std::move(entries));
     });
   });
+  if (!accepted) {
+    loop_->post([cb] { cb(false, {}); });
+  }
 }
 
 bool ConsulBackend::compareAndSwap(const std::string& service,

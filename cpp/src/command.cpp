@@ -8,6 +8,7 @@
 
 #include <cctype>
 #include <cstring>
+#include <map>
 
 #include "cpilot/log.hpp"
 #include "cpilot/spawner.hpp"
@@ -15,6 +16,32 @@
 extern char** environ;
 
 namespace cpilot {
+
+namespace {
+
+// CONTAINERPILOT_{NAME}_PID vars for running execs. Kept as an overlay
+// merged into each spawn's env snapshot instead of calling setenv per
+// exec: glibc's setenv leaks the replaced "NAME=value" string on every
+// update (it may still be referenced), which at ~1000 execs/sec grew
+// the daemon ~70 KB/s. Loop-thread only.
+std::map<std::string, std::string>& pidEnvOverlay() {
+  static std::map<std::string, std::string> overlay;
+  return overlay;
+}
+
+std::vector<std::string> snapshotEnvWithOverlay() {
+  auto& overlay = pidEnvOverlay();
+  std::vector<std::string> env;
+  for (char** e = ::environ; *e; e++) {
+    const char* eq = strchr(*e, '=');
+    if (eq && overlay.count(std::string(*e, eq - *e))) continue;
+    env.emplace_back(*e);
+  }
+  for (auto& kv : overlay) env.push_back(kv.first + "=" + kv.second);
+  return env;
+}
+
+}  // namespace
 
 bool parseArgs(const Json& raw, std::string* execPath,
                std::vector<std::string>* args, std::string* err) {
@@ -126,10 +153,9 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   pendingSignal_ = 0;
   auto self = shared_from_this();
   int readFd = pipefds[0];
-  // snapshot env on this (the loop) thread; the spawner must not read
-  // the live environ concurrently with our setenv calls
-  std::vector<std::string> envCopy;
-  for (char** e = ::environ; *e; e++) envCopy.emplace_back(*e);
+  // snapshot env (+ the PID-var overlay) on this (the loop) thread; the
+  // spawner must never read the live environ concurrently with setenv
+  std::vector<std::string> envCopy = snapshotEnvWithOverlay();
   // the spawner thread does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
   Spawner::global().spawn(
@@ -148,7 +174,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
         pid_ = pid;
         if (pidEnvName_.empty())
           pidEnvName_ = "CONTAINERPILOT_" + envName() + "_PID";
-        setenv(pidEnvName_.c_str(), std::to_string(pid).c_str(), 1);
+        pidEnvOverlay()[pidEnvName_] = std::to_string(pid);
 
         if (!raw_) {
           logFd_ = readFd;
@@ -233,7 +259,7 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
     logFd_ = -1;
   }
 
-  unsetenv(pidEnvName_.c_str());
+  if (!pidEnvName_.empty()) pidEnvOverlay().erase(pidEnvName_);
 
   running_ = false;
   pid_ = -1;

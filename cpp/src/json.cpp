@@ -115,7 +115,17 @@ class Parser {
         fail("expected ':' after object key");
       }
       skipWs();
-      obj.emplace_back(std::move(key), parseValue());
+      Json value = parseValue();
+      // duplicate keys: last one wins (Go map unmarshal semantics)
+      bool replaced = false;
+      for (auto& kv : obj) {
+        if (kv.first == key) {
+          kv.second = std::move(value);
+          replaced = true;
+          break;
+        }
+      }
+      if (!replaced) obj.emplace_back(std::move(key), std::move(value));
       skipWs();
       char c = get();
       if (c == ',') {

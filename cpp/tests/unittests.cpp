@@ -85,6 +85,9 @@ static void testJson5() {
   // serializer round-trip
   CHECK_EQ(parseJson5("{\"a\":[1,\"x\"],\"b\":true}").dump(),
            std::string("{\"a\":[1,\"x\"],\"b\":true}"));
+
+  // duplicate keys: last wins (Go map unmarshal semantics)
+  CHECK_EQ(parseJson5("{a: 1, a: 2}").find("a")->asInt(), (int64_t)2);
 }
 
 static void testDurations() {

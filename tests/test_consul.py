@@ -186,3 +186,30 @@ def test_watch_tag_and_dc_in_query(daemon_factory, mock_consul):
         for _, path in mock_consul.requests)), mock_consul.requests[-5:]
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_job_gated_on_watch_healthy(daemon_factory, mock_consul):
+    """A job with when: {source: watch.db, once: healthy} starts only
+    after the upstream becomes available — the canonical 'wait for the
+    database' pattern."""
+    mock_consul.set_health("db", [])
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "logging": {"level": "DEBUG"},
+        "jobs": [
+            {"name": "main-app", "exec": "sleep 60"},
+            {"name": "migrate", "exec": "echo migrated",
+             "when": {"source": "watch.db", "once": "healthy"}},
+        ],
+        "watches": [{"name": "db", "interval": 1}],
+    }).start()
+    d.wait_for_socket()
+    time.sleep(1.5)
+    assert "migrated" not in d.log()
+    mock_consul.set_health("db", [
+        {"ID": "db-1", "Address": "10.0.0.9", "Port": 5432}])
+    assert wait_until(lambda: "migrated" in d.log()), d.log()
+    assert d.log().count("migrated") == 1
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -169,3 +169,39 @@ def test_subcommands_over_socket(daemon_factory):
     d.wait_for_socket()
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_putenv_then_reload_rerenders_template(daemon_factory, tmp_path):
+    """The canonical /v3/environ workflow: set an env var, reload, and
+    the re-rendered config template picks it up (templates render
+    against the daemon's current environment)."""
+    d = daemon_factory(config_text="""
+{
+  consul: "localhost:79",
+  stopTimeout: 1,
+  logging: {level: "DEBUG"},
+  control: {socket: "{SOCKET}"},
+  jobs: [
+    {name: "main-app", "exec": "sleep 60"},
+    {name: "versionprinter",
+     exec: "echo version={{ .DEPLOY_VERSION | default "none" }}"},
+  ],
+}
+""").start()
+    d.wait_for_socket()
+    deadline = time.time() + 10
+    while time.time() < deadline and "version=none" not in d.log():
+        time.sleep(0.1)
+    assert "version=none" in d.log()
+
+    status, _ = d.control("POST", "/v3/environ",
+                          json.dumps({"DEPLOY_VERSION": "v2.5"}))
+    assert status == 200
+    status, _ = d.control("POST", "/v3/reload")
+    assert status == 200
+    deadline = time.time() + 15
+    while time.time() < deadline and "version=v2.5" not in d.log():
+        time.sleep(0.1)
+    assert "version=v2.5" in d.log()
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -24,6 +24,8 @@ class Command : public std::enable_shared_from_this<Command> {
  public:
   Command(std::string execPath, std::vector<std::string> args, Duration timeout,
           bool raw, std::string logField);
+  Command(const Command&) = delete;
+  Command& operator=(const Command&) = delete;
 
   const std::string& name() const { return name_; }
   void setName(std::string name) { name_ = std::move(name); }

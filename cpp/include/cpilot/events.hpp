@@ -76,6 +76,8 @@ class Subscriber {
 class Bus : public std::enable_shared_from_this<Bus> {
  public:
   explicit Bus(Loop& loop);
+  Bus(const Bus&) = delete;
+  Bus& operator=(const Bus&) = delete;
 
   void subscribe(Subscriber* s);
   void unsubscribe(Subscriber* s);

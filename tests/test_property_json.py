@@ -66,3 +66,14 @@ def test_parser_never_crashes_on_garbage(text):
         native.json5_to_json(text)
     except ValueError:
         pass  # rejecting is fine; crashing is not
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.text(alphabet=st.sampled_from(list("{}().|$\"ab \n-")), max_size=50))
+def test_template_never_crashes_on_garbage(text):
+    """The template engine either renders or raises ValueError — never
+    crashes the process."""
+    try:
+        native.render_template(text)
+    except ValueError:
+        pass

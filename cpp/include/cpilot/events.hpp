@@ -71,6 +71,13 @@ class Subscriber {
  public:
   virtual ~Subscriber() = default;
   virtual void onEvent(const Event& event) = 0;
+  // hot-path variant: srcHash = std::hash of event.source, computed once
+  // per event by the bus so high-fan-out subscribers can compare a
+  // precomputed hash instead of strings (default forwards)
+  virtual void onEventHashed(const Event& event, size_t srcHash) {
+    (void)srcHash;
+    onEvent(event);
+  }
 };
 
 class Bus : public std::enable_shared_from_this<Bus> {

@@ -96,6 +96,7 @@ class Job : public Subscriber, public std::enable_shared_from_this<Job> {
   void kill();
 
   void onEvent(const Event& event) override;
+  void onEventHashed(const Event& event, size_t srcHash) override;
 
  private:
   enum class Phase { Running, StoppingWait, Complete };
@@ -103,7 +104,8 @@ class Job : public Subscriber, public std::enable_shared_from_this<Job> {
   static constexpr bool kContinue = false, kHalt = true;
 
   void processEvent(const Event& event);
-  HandleResult dispatch(const Event& event);
+  void processEventHashed(const Event& event, size_t srcHash);
+  HandleResult dispatch(const Event& event, size_t srcHash);
   HandleResult onHeartbeatTimerExpired();
   HandleResult onStartTimeoutExpired();
   HandleResult onRunEveryTimerExpired();
@@ -124,10 +126,12 @@ class Job : public Subscriber, public std::enable_shared_from_this<Job> {
   void finishCleanup();
 
   std::string name_;
-  // precomputed event-match sources (hot path: every bus event hits
-  // dispatch() in every job)
+  // precomputed event-match sources + their hashes (hot path: every bus
+  // event hits dispatch() in every job; hashes keep the non-matching
+  // common case to one cache line instead of string compares)
   std::string heartbeatSource_, runEverySource_, healthCheckName_,
       stoppingTimeoutSource_;
+  size_t hName_ = 0, hHeartbeat_ = 0, hRunEvery_ = 0, hCheck_ = 0;
   CommandPtr exec_;
   JobStatus status_ = JobStatus::Idle;
   std::shared_ptr<ServiceDefinition> service_;

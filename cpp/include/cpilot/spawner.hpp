@@ -1,7 +1,8 @@
 // Dedicated spawner thread: posix_spawnp runs off the reactor so a burst
 // of health-check launches never delays event dispatch. Spawn completions
 // are posted back onto the loop; Loop::watchChild handles the
-// SIGCHLD-before-completion race via its unclaimed-exit map.
+// SIGCHLD-before-completion race via its unclaimed-exit map. See the
+// constructor comment for why this is exactly one thread.
 #pragma once
 
 #include <sys/types.h>
@@ -45,7 +46,7 @@ class Spawner {
     SpawnCb cb;
   };
 
-  std::thread thread_;
+  std::vector<std::thread> threads_;
   std::mutex mu_;
   std::condition_variable cv_;
   std::deque<Request> queue_;

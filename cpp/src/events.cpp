@@ -132,13 +132,14 @@ void Bus::drain() {
     // their slot (checked per delivery), new subscribes append past the
     // bound captured here so they don't see this event
     size_t bound = subscribers_.size();
+    size_t srcHash = std::hash<std::string>{}(event.source);
     uint64_t batch = 0;
     for (size_t si = 0; si < bound; si++) {
       Subscriber* s = subscribers_[si];
       if (!s) continue;
       delivered_++;
       batch++;
-      s->onEvent(event);
+      s->onEventHashed(event, srcHash);
     }
     if (batch) deliveriesCounter_->inc({}, (double)batch);
   }

@@ -516,6 +516,11 @@ Job::Job(const std::shared_ptr<JobConfig>& cfg)
   healthCheckName_ =
       healthCheckExec_ ? healthCheckExec_->name() : ("check." + name_);
   stoppingTimeoutSource_ = name_ + ".stopping-timeout";
+  std::hash<std::string> h;
+  hName_ = h(name_);
+  hHeartbeat_ = h(heartbeatSource_);
+  hRunEvery_ = h(runEverySource_);
+  hCheck_ = h(healthCheckName_);
   if (name_ == "containerpilot") {
     // hardcoded always-healthy telemetry job (jobs/jobs.go:82-87)
     status_ = JobStatus::AlwaysHealthy;
@@ -571,9 +576,19 @@ void Job::kill() {
   if (exec_) exec_->kill();
 }
 
-void Job::onEvent(const Event& event) { processEvent(event); }
+void Job::onEvent(const Event& event) {
+  processEventHashed(event, std::hash<std::string>{}(event.source));
+}
+
+void Job::onEventHashed(const Event& event, size_t srcHash) {
+  processEventHashed(event, srcHash);
+}
 
 void Job::processEvent(const Event& event) {
+  processEventHashed(event, std::hash<std::string>{}(event.source));
+}
+
+void Job::processEventHashed(const Event& event, size_t srcHash) {
   if (phase_ == Phase::Complete) return;
   // registration retry inside the event loop (jobs/jobs.go:168-171)
   checkRegistration();
@@ -593,29 +608,34 @@ void Job::processEvent(const Event& event) {
     cleanup();
     return;
   }
-  if (dispatch(event) == kHalt) cleanup();
+  if (dispatch(event, srcHash) == kHalt) cleanup();
 }
 
-Job::HandleResult Job::dispatch(const Event& event) {
+Job::HandleResult Job::dispatch(const Event& event, size_t srcHash) {
   // match order mirrors the reference switch (jobs/jobs.go:195-232);
-  // code compared before source so non-matching events exit cheaply
+  // code then source-hash compared first so the common non-matching
+  // case never touches the string buffers
+  auto srcIs = [&](size_t h, const std::string& s) {
+    return srcHash == h && event.source == s;
+  };
   switch (event.code) {
     case EventCode::TimerExpired:
-      if (event.source == heartbeatSource_) return onHeartbeatTimerExpired();
+      if (srcIs(hHeartbeat_, heartbeatSource_))
+        return onHeartbeatTimerExpired();
       if (startTimeoutEvent_ != NonEvent && event == startTimeoutEvent_)
         return onStartTimeoutExpired();
-      if (event.source == runEverySource_) return onRunEveryTimerExpired();
+      if (srcIs(hRunEvery_, runEverySource_)) return onRunEveryTimerExpired();
       break;
     case EventCode::ExitFailed:
-      if (event.source == healthCheckName_) return onHealthCheckFailed();
-      if (event.source == name_) return onExecExit();
+      if (srcIs(hCheck_, healthCheckName_)) return onHealthCheckFailed();
+      if (srcIs(hName_, name_)) return onExecExit();
       break;
     case EventCode::ExitSuccess:
-      if (event.source == healthCheckName_) return onHealthCheckPassed();
-      if (event.source == name_) return onExecExit();
+      if (srcIs(hCheck_, healthCheckName_)) return onHealthCheckPassed();
+      if (srcIs(hName_, name_)) return onExecExit();
       break;
     case EventCode::Quit:
-      if (event.source == name_) return onQuit();
+      if (srcIs(hName_, name_)) return onQuit();
       break;
     case EventCode::Shutdown:
       if (event == GlobalShutdown) return onQuit();

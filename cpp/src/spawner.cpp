@@ -16,8 +16,14 @@ Spawner& Spawner::global() {
 }
 
 Spawner::Spawner() {
-  thread_ = std::thread([this] { threadMain(); });
-  thread_.detach();  // process-lifetime singleton
+  // deliberately ONE thread: it serializes launches, which acts as
+  // natural backpressure past saturation. A 4-thread pool was measured
+  // to push ~2x the launch rate at 5x baseline load but let completion
+  // bursts flood the reactor (p99 dispatch 1 ms -> 50+ ms); a supervisor
+  // should shed overlapping checks (reference single-instance exec
+  // semantics) rather than trade latency for past-saturation throughput.
+  threads_.emplace_back([this] { threadMain(); });
+  threads_.back().detach();  // process-lifetime singleton
 }
 
 void Spawner::spawn(Loop& loop, std::string execPath,

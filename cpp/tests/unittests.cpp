@@ -547,6 +547,70 @@ static void testWhenEachFiresRepeatedly() {
   CHECK_EQ(probe.count(EventCode::ExitSuccess, "onchange"), 3);
 }
 
+static void testMaintenanceMatrix() {
+  // maintenance-mode matrix (jobs/jobs_test.go:208-269): health events
+  // are suppressed during maintenance and resume after
+  JobScenario sc;
+  Probe probe;
+  sc.bus->subscribe(&probe);
+  sc.addJob(R"({name: "svc", exec: "sleep 10",
+                health: {exec: "true", interval: "200ms", ttl: 5}})");
+  sc.loop.addTimeout(std::chrono::milliseconds(600), [&sc] {
+    sc.bus->publish(GlobalEnterMaintenance);
+  });
+  sc.loop.addTimeout(std::chrono::milliseconds(1400), [&sc] {
+    sc.bus->publish(GlobalExitMaintenance);
+  });
+  int healthyAtEnter = -1, healthyAtExit = -1;
+  sc.loop.addTimeout(std::chrono::milliseconds(1300), [&] {
+    healthyAtEnter = probe.count(EventCode::StatusHealthy, "svc");
+  });
+  sc.loop.addTimeout(std::chrono::milliseconds(1450), [&] {
+    healthyAtExit = probe.count(EventCode::StatusHealthy, "svc");
+  });
+  sc.run(2200);
+  // some health events before maintenance
+  CHECK(healthyAtEnter >= 1);
+  // no healthy events published during the maintenance window
+  CHECK_EQ(healthyAtExit, healthyAtEnter);
+  // and they resume after exitMaintenance
+  CHECK(probe.count(EventCode::StatusHealthy, "svc") > healthyAtExit);
+}
+
+static void testSignalJobMatrix() {
+  // signal-driven jobs fire on their own signal only and repeatedly
+  // (jobs/jobs.go:351-357)
+  JobScenario sc;
+  Probe probe;
+  sc.bus->subscribe(&probe);
+  sc.addJob(R"({name: "onhup", exec: "true", when: {source: "SIGHUP"}})");
+  sc.addJob(R"({name: "onusr2", exec: "true", when: {source: "SIGUSR2"}})");
+  for (int i = 1; i <= 2; i++) {
+    sc.loop.addTimeout(std::chrono::milliseconds(200 * i),
+                       [&sc] { sc.bus->publishSignal("SIGHUP"); });
+  }
+  sc.loop.addTimeout(std::chrono::milliseconds(500),
+                     [&sc] { sc.bus->publishSignal("SIGUSR2"); });
+  sc.run(1300);
+  CHECK_EQ(probe.count(EventCode::ExitSuccess, "onhup"), 2);
+  CHECK_EQ(probe.count(EventCode::ExitSuccess, "onusr2"), 1);
+}
+
+static void testStartTimeoutQuitsJob() {
+  // when.timeout fires -> TimerExpired{job} published + job quits
+  // (jobs/jobs.go:259-264)
+  JobScenario sc;
+  Probe probe;
+  sc.bus->subscribe(&probe);
+  sc.addJob(R"({name: "patient", exec: "true",
+                when: {source: "ghost", once: "exitSuccess",
+                       timeout: "300ms"}})");
+  sc.run(1500);
+  CHECK_EQ(probe.count(EventCode::ExitSuccess, "patient"), 0);
+  CHECK_EQ(probe.count(EventCode::TimerExpired, "patient"), 1);
+  CHECK_EQ(probe.count(EventCode::Stopped, "patient"), 1);
+}
+
 int main() {
   testJson5();
   testDurations();
@@ -561,6 +625,9 @@ int main() {
   testPeriodicJobRuns();
   testPreStopJobOnShutdown();
   testWhenEachFiresRepeatedly();
+  testMaintenanceMatrix();
+  testSignalJobMatrix();
+  testStartTimeoutQuitsJob();
   if (failures) {
     fprintf(stderr, "%d failures\n", failures);
     return 1;

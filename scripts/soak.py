@@ -1,4 +1,5 @@
-"""5-minute stress soak with reloads: RSS/CPU stability evidence."""
+"""Stress soak with reloads + maintenance toggles: RSS/CPU stability
+evidence. Duration via argv[1] seconds (default 300)."""
 import json, os, sys, tempfile, time
 sys.path.insert(0, os.getcwd())
 from containerpilot_amd import harness
@@ -22,25 +23,34 @@ def sample(pid):
     cpu = (int(parts[13]) + int(parts[14])) / os.sysconf("SC_CLK_TCK")
     return int(rss), cpu
 
+duration = int(sys.argv[1]) if len(sys.argv) > 1 else 300
 samples = []
 t0 = time.time()
 last_cpu = sample(d.proc.pid)[1]
 reloads = 0
-while time.time() - t0 < 300:
+maint = 0
+while time.time() - t0 < duration:
     time.sleep(10)
     rss, cpu = sample(d.proc.pid)
     samples.append({"t": round(time.time()-t0), "rss_kb": rss,
                     "cpu_pct": round((cpu-last_cpu)/10*100, 1)})
     last_cpu = cpu
-    if len(samples) in (10, 20):  # reload at ~100s and ~200s
+    n = len(samples)
+    if n % 10 == 0:  # reload every ~100s
         d.control("POST", "/v3/reload")
         reloads += 1
         time.sleep(1); d.wait_for_socket(timeout=20)
+    elif n % 10 == 4:  # maintenance window every ~100s
+        d.control("POST", "/v3/maintenance/enable")
+        maint += 1
+    elif n % 10 == 6:
+        d.control("POST", "/v3/maintenance/disable")
 
 alive = d.proc.poll() is None
 d.terminate(); rc = d.wait(timeout=60)
 mc.stop()
-out = {"duration_s": 300, "reloads": reloads, "alive_throughout": alive,
+out = {"duration_s": duration, "reloads": reloads,
+       "maintenance_windows": maint, "alive_throughout": alive,
        "clean_exit_rc": rc,
        "rss_kb_first": samples[0]["rss_kb"], "rss_kb_last": samples[-1]["rss_kb"],
        "rss_kb_max": max(s["rss_kb"] for s in samples),

@@ -8,6 +8,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <random>
 #include <cstring>
 #include <thread>
 
@@ -194,6 +195,14 @@ void App::teardownGeneration() {
   totalDelivered_ += bus_->deliveredCount();
   const auto& window = bus_->latencyWindow();
   latencies_.insert(latencies_.end(), window.begin(), window.end());
+  // bound the whole-run stats sample across many reload generations
+  // (each generation contributes up to 64k samples)
+  constexpr size_t kMaxSamples = 262144;
+  if (latencies_.size() > kMaxSamples) {
+    std::minstd_rand rng(42);
+    std::shuffle(latencies_.begin(), latencies_.end(), rng);
+    latencies_.resize(kMaxSamples / 2);
+  }
 
   loop_.stop();
 }

@@ -547,6 +547,21 @@ static void testWhenEachFiresRepeatedly() {
   CHECK_EQ(probe.count(EventCode::ExitSuccess, "onchange"), 3);
 }
 
+static void testDebugRing() {
+  // the bus's 10-slot circular debug buffer (events/bus.go:24-54):
+  // keeps the last 10 events, drains oldest-first
+  Loop loop;
+  auto bus = std::make_shared<Bus>(loop);
+  for (int i = 0; i < 13; i++)
+    bus->publish(Event{EventCode::Metric, "ev-" + std::to_string(i)});
+  auto drained = bus->debugEvents();
+  CHECK_EQ(drained.size(), (size_t)10);
+  CHECK_EQ(drained.front().source, std::string("ev-3"));
+  CHECK_EQ(drained.back().source, std::string("ev-12"));
+  // drained: a second read is empty
+  CHECK_EQ(bus->debugEvents().size(), (size_t)0);
+}
+
 static void testMaintenanceMatrix() {
   // maintenance-mode matrix (jobs/jobs_test.go:208-269): health events
   // are suppressed during maintenance and resume after
@@ -625,6 +640,7 @@ int main() {
   testPeriodicJobRuns();
   testPreStopJobOnShutdown();
   testWhenEachFiresRepeatedly();
+  testDebugRing();
   testMaintenanceMatrix();
   testSignalJobMatrix();
   testStartTimeoutQuitsJob();

@@ -213,3 +213,37 @@ def test_job_gated_on_watch_healthy(daemon_factory, mock_consul):
     assert d.log().count("migrated") == 1
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_watch_tag_filters_instances(daemon_factory, mock_consul):
+    """Only instances carrying the watch's tag count as upstream members
+    (tag-partitioned pools)."""
+    mock_consul.set_health("pool", [
+        {"ID": "pool-1", "Address": "10.0.0.1", "Port": 80,
+         "Tags": ["blue"]},
+        {"ID": "pool-2", "Address": "10.0.0.2", "Port": 80,
+         "Tags": ["green"]},
+    ])
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "logging": {"level": "DEBUG"},
+        "jobs": [{"name": "main-app", "exec": "sleep 60"}],
+        "watches": [{"name": "pool", "interval": 1, "tag": "blue"}],
+    }).start()
+    d.wait_for_socket()
+    # becomes healthy (one blue instance)
+    assert wait_until(lambda: "{StatusHealthy watch.pool}" in d.log())
+    marker = len(d.log())
+    # removing the green instance is invisible to a blue-tag watch
+    mock_consul.set_health("pool", [
+        {"ID": "pool-1", "Address": "10.0.0.1", "Port": 80,
+         "Tags": ["blue"]},
+    ])
+    time.sleep(2.5)
+    assert "{StatusChanged watch.pool}" not in d.log()[marker:]
+    # removing the blue instance fires unhealthy
+    mock_consul.set_health("pool", [])
+    assert wait_until(lambda: "{StatusUnhealthy watch.pool}" in d.log())
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -608,6 +608,75 @@ Value fnLoop(const std::vector<Value>& args) {
   return Value::mklist(out);
 }
 
+// ---------- Go text/template builtins (always available in Go) ----------
+
+bool valueEq(const Value& a, const Value& b) {
+  bool aNum = a.kind == Value::Kind::Int || a.kind == Value::Kind::Float;
+  bool bNum = b.kind == Value::Kind::Int || b.kind == Value::Kind::Float;
+  if (aNum && bNum) {
+    double av = a.kind == Value::Kind::Int ? (double)a.i : a.f;
+    double bv = b.kind == Value::Kind::Int ? (double)b.i : b.f;
+    return av == bv;
+  }
+  if (a.kind == Value::Kind::Bool || b.kind == Value::Kind::Bool)
+    return a.truthy() == b.truthy();
+  return a.print() == b.print();
+}
+
+int valueCmp(const Value& a, const Value& b) {
+  bool aNum = a.kind == Value::Kind::Int || a.kind == Value::Kind::Float;
+  bool bNum = b.kind == Value::Kind::Int || b.kind == Value::Kind::Float;
+  if (aNum && bNum) {
+    double av = a.kind == Value::Kind::Int ? (double)a.i : a.f;
+    double bv = b.kind == Value::Kind::Int ? (double)b.i : b.f;
+    return av < bv ? -1 : (av > bv ? 1 : 0);
+  }
+  return a.print().compare(b.print()) < 0
+             ? -1
+             : (a.print() == b.print() ? 0 : 1);
+}
+
+Value fnLen(const std::vector<Value>& args) {
+  if (args.size() != 1)
+    throw std::runtime_error("template: wrong number of args for len");
+  const Value& v = args[0];
+  if (v.kind == Value::Kind::List) return Value::integer((int64_t)v.list->size());
+  if (v.kind == Value::Kind::Str) return Value::integer((int64_t)v.s.size());
+  throw std::runtime_error("template: len of unsupported type");
+}
+
+Value fnIndex(const std::vector<Value>& args) {
+  if (args.size() != 2)
+    throw std::runtime_error("template: wrong number of args for index");
+  int64_t i = args[1].toInt();
+  if (args[0].kind == Value::Kind::List) {
+    if (i < 0 || (size_t)i >= args[0].list->size())
+      throw std::runtime_error("template: index out of range");
+    return (*args[0].list)[i];
+  }
+  if (args[0].kind == Value::Kind::Str) {
+    if (i < 0 || (size_t)i >= args[0].s.size())
+      throw std::runtime_error("template: index out of range");
+    return Value::integer((unsigned char)args[0].s[i]);
+  }
+  throw std::runtime_error("template: index of unsupported type");
+}
+
+Value fnPrint(const std::vector<Value>& args, bool newline) {
+  // fmt.Sprint semantics: spaces between operands when neither is a string
+  std::string out;
+  for (size_t i = 0; i < args.size(); i++) {
+    if (i > 0 && args[i - 1].kind != Value::Kind::Str &&
+        args[i].kind != Value::Kind::Str && !newline)
+      out += " ";
+    else if (i > 0 && newline)
+      out += " ";  // Sprintln: always spaces
+    out += args[i].print();
+  }
+  if (newline) out += "\n";
+  return Value::str(out);
+}
+
 Value fnPrintf(const std::vector<Value>& args) {
   if (args.empty())
     throw std::runtime_error("template: printf needs a format string");
@@ -773,6 +842,53 @@ class Evaluator {
     if (name == "regexReplaceAll") return fnRegexReplaceAll(args);
     if (name == "loop") return fnLoop(args);
     if (name == "printf") return fnPrintf(args);
+    // Go text/template builtins (template.go's FuncMap only ADDS funcs;
+    // the language builtins are always available to reference configs)
+    if (name == "eq") {
+      if (args.size() < 2)
+        throw std::runtime_error("template: wrong number of args for eq");
+      for (size_t i = 1; i < args.size(); i++)
+        if (valueEq(args[0], args[i])) return Value::boolean(true);
+      return Value::boolean(false);
+    }
+    if (name == "ne") {
+      if (args.size() != 2)
+        throw std::runtime_error("template: wrong number of args for ne");
+      return Value::boolean(!valueEq(args[0], args[1]));
+    }
+    if (name == "lt" || name == "le" || name == "gt" || name == "ge") {
+      if (args.size() != 2)
+        throw std::runtime_error("template: wrong number of args for " + name);
+      int c = valueCmp(args[0], args[1]);
+      bool r = (name == "lt")   ? c < 0
+               : (name == "le") ? c <= 0
+               : (name == "gt") ? c > 0
+                                : c >= 0;
+      return Value::boolean(r);
+    }
+    if (name == "and") {
+      if (args.empty())
+        throw std::runtime_error("template: wrong number of args for and");
+      for (auto& a : args)
+        if (!a.truthy()) return a;
+      return args.back();
+    }
+    if (name == "or") {
+      if (args.empty())
+        throw std::runtime_error("template: wrong number of args for or");
+      for (auto& a : args)
+        if (a.truthy()) return a;
+      return args.back();
+    }
+    if (name == "not") {
+      if (args.size() != 1)
+        throw std::runtime_error("template: wrong number of args for not");
+      return Value::boolean(!args[0].truthy());
+    }
+    if (name == "len") return fnLen(args);
+    if (name == "index") return fnIndex(args);
+    if (name == "print") return fnPrint(args, false);
+    if (name == "println") return fnPrint(args, true);
     throw std::runtime_error("template: function \"" + name + "\" not defined");
   }
 };

@@ -175,6 +175,25 @@ static void testTemplate() {
   CHECK_EQ(renderTemplate("{{ range $i := loop 2 5 -}}i={{$i}},{{ end }}"),
            std::string("i=2,i=3,i=4,"));
   unsetenv("COUNT");
+
+  // Go template builtins are available to reference configs
+  setenv("STAGE", "prod", 1);
+  CHECK_EQ(renderTemplate("{{ if eq .STAGE \"prod\" }}P{{ else }}D{{ end }}"),
+           std::string("P"));
+  CHECK_EQ(renderTemplate("{{ if ne .STAGE \"dev\" }}y{{ end }}"),
+           std::string("y"));
+  CHECK_EQ(renderTemplate("{{ if and .STAGE .TEST_NAME }}both{{ end }}"),
+           std::string("both"));
+  CHECK_EQ(renderTemplate("{{ or .TEST_MISSING \"fallback\" }}"),
+           std::string("fallback"));
+  CHECK_EQ(renderTemplate("{{ if not .TEST_MISSING }}empty{{ end }}"),
+           std::string("empty"));
+  CHECK_EQ(renderTemplate("{{ len \"abcd\" }}"), std::string("4"));
+  CHECK_EQ(renderTemplate("{{ len (loop 3) }}"), std::string("3"));
+  CHECK_EQ(renderTemplate("{{ index (loop 5 8) 1 }}"), std::string("6"));
+  CHECK_EQ(renderTemplate("{{ if lt 1 2 }}y{{ end }}"), std::string("y"));
+  CHECK_EQ(renderTemplate("{{ print \"a\" 1 2 }}"), std::string("a1 2"));
+  unsetenv("STAGE");
 }
 
 static void testEvents() {

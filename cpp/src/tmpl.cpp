@@ -135,18 +135,20 @@ struct Node;
 using NodePtr = std::shared_ptr<Node>;
 
 struct Node {
-  enum class Kind { Text, Action, If, Range } kind;
+  enum class Kind { Text, Action, If, Range, Assign, With } kind;
   std::string text;               // Text
-  ExprPtr expr;                   // Action/If/Range condition
-  std::string rangeVar;           // Range: $i
-  std::vector<NodePtr> body;      // If/Range
-  std::vector<NodePtr> elseBody;  // If
+  ExprPtr expr;                   // Action/If/Range/With/Assign rhs
+  std::string rangeVar;           // Range: $i (index var) / Assign: name
+  std::string rangeVar2;          // Range: $v (value var, two-var form)
+  std::vector<NodePtr> body;      // If/Range/With
+  std::vector<NodePtr> elseBody;  // If/With
 };
 
 // ---------- lexer for actions ----------
 
 struct ActionTok {
-  enum class Kind { Ident, Field, Var, Str, Num, LParen, RParen, Pipe, Assign, End } kind;
+  enum class Kind { Ident, Field, Var, Str, Num, LParen, RParen, Pipe, Assign,
+                    Comma, End } kind;
   std::string text;
   Value num;
 };
@@ -162,9 +164,14 @@ class ActionLexer {
     if (c == '(') { pos_++; return {ActionTok::Kind::LParen, "(", {}}; }
     if (c == ')') { pos_++; return {ActionTok::Kind::RParen, ")", {}}; }
     if (c == '|') { pos_++; return {ActionTok::Kind::Pipe, "|", {}}; }
+    if (c == ',') { pos_++; return {ActionTok::Kind::Comma, ",", {}}; }
     if (c == ':' && pos_ + 1 < s_.size() && s_[pos_ + 1] == '=') {
       pos_ += 2;
       return {ActionTok::Kind::Assign, ":=", {}};
+    }
+    if (c == '=') {
+      pos_++;
+      return {ActionTok::Kind::Assign, "=", {}};
     }
     if (c == '"' || c == '\'' || c == '`') return lexString(c);
     if (c == '.') {
@@ -281,18 +288,40 @@ class ActionParser {
 
   bool atEnd() const { return tok_.kind == ActionTok::Kind::End; }
 
-  // parse "range $i := pipeline" header; returns varname ("" if none)
-  std::string parseRangeVar() {
-    if (tok_.kind == ActionTok::Kind::Var) {
-      std::string name = tok_.text;
+  // parse "range $i := ..." / "range $i, $v := ..." header; returns
+  // (var1, var2) — empty strings when absent
+  std::pair<std::string, std::string> parseRangeVars() {
+    std::pair<std::string, std::string> out;
+    if (tok_.kind != ActionTok::Kind::Var) return out;
+    out.first = tok_.text;
+    advance();
+    if (tok_.kind == ActionTok::Kind::Comma) {
       advance();
-      if (tok_.kind == ActionTok::Kind::Assign) {
-        advance();
-        return name;
-      }
-      throw std::runtime_error("template: expected := after range variable");
+      if (tok_.kind != ActionTok::Kind::Var)
+        throw std::runtime_error("template: expected variable after ','");
+      out.second = tok_.text;
+      advance();
     }
-    return "";
+    if (tok_.kind != ActionTok::Kind::Assign)
+      throw std::runtime_error("template: expected := after range variable");
+    advance();
+    return out;
+  }
+
+  // "$x := pipeline" as a plain action; returns (name, expr) or ("",null)
+  std::pair<std::string, ExprPtr> tryParseAssignment() {
+    if (tok_.kind != ActionTok::Kind::Var) return {"", nullptr};
+    std::string name = tok_.text;
+    ActionLexer save = lex_;
+    ActionTok savedTok = tok_;
+    advance();
+    if (tok_.kind == ActionTok::Kind::Assign) {
+      advance();
+      return {name, parsePipeline()};
+    }
+    lex_ = save;
+    tok_ = savedTok;
+    return {"", nullptr};
   }
 
  private:
@@ -412,34 +441,30 @@ class TemplateParser {
       pendingTrim_ = trimR;
 
       std::string trimmed = trim(body);
+      // template comments: {{/* ... */}}
+      if (trimmed.rfind("/*", 0) == 0 &&
+          trimmed.size() >= 4 &&
+          trimmed.compare(trimmed.size() - 2, 2, "*/") == 0) {
+        continue;
+      }
       std::string keyword = firstWord(trimmed);
       if (keyword == "end" || keyword == "else") {
         if (!terminator)
           throw std::runtime_error("template: unexpected {{" + keyword + "}}");
-        *terminator = keyword;
+        *terminator = trimmed;  // full body: "else if ..." keeps its tail
         return nodes;
       }
-      if (keyword == "if") {
-        auto node = std::make_shared<Node>();
-        node->kind = Node::Kind::If;
-        ActionParser ap(trimmed.substr(2));
-        node->expr = ap.parsePipeline();
-        std::string term;
-        node->body = parseNodes(&term);
-        if (term == "else") {
-          std::string term2;
-          node->elseBody = parseNodes(&term2);
-          if (term2 != "end")
-            throw std::runtime_error("template: expected {{end}}");
-        }
-        nodes.push_back(node);
+      if (keyword == "if" || keyword == "with") {
+        nodes.push_back(parseIfChain(trimmed));
         continue;
       }
       if (keyword == "range") {
         auto node = std::make_shared<Node>();
         node->kind = Node::Kind::Range;
         ActionParser ap(trimmed.substr(5));
-        node->rangeVar = ap.parseRangeVar();
+        auto vars = ap.parseRangeVars();
+        node->rangeVar = vars.first;
+        node->rangeVar2 = vars.second;
         node->expr = ap.parsePipeline();
         std::string term;
         node->body = parseNodes(&term);
@@ -447,6 +472,19 @@ class TemplateParser {
           throw std::runtime_error("template: expected {{end}} to close range");
         nodes.push_back(node);
         continue;
+      }
+      // assignment action: {{ $x := expr }}
+      {
+        ActionParser ap(trimmed);
+        auto assign = ap.tryParseAssignment();
+        if (!assign.first.empty()) {
+          auto node = std::make_shared<Node>();
+          node->kind = Node::Kind::Assign;
+          node->rangeVar = assign.first;
+          node->expr = assign.second;
+          nodes.push_back(node);
+          continue;
+        }
       }
       // plain action
       auto node = std::make_shared<Node>();
@@ -458,6 +496,35 @@ class TemplateParser {
     if (terminator && pos_ >= text_.size() && terminator->empty())
       throw std::runtime_error("template: unexpected EOF, expected {{end}}");
     return nodes;
+  }
+
+  // parse "{{if expr}}" or "{{with expr}}" including else / else-if
+  // chains, consuming exactly one {{end}}
+  NodePtr parseIfChain(const std::string& header) {
+    bool isWith = firstWord(header) == "with";
+    auto node = std::make_shared<Node>();
+    node->kind = isWith ? Node::Kind::With : Node::Kind::If;
+    ActionParser ap(header.substr(isWith ? 4 : 2));
+    node->expr = ap.parsePipeline();
+    std::string term;
+    node->body = parseNodes(&term);
+    if (firstWord(term) == "else") {
+      std::string rest = trim(term.substr(4));
+      if (!rest.empty()) {
+        // {{else if ...}}: nested chain shares our {{end}}
+        if (firstWord(rest) != "if")
+          throw std::runtime_error("template: expected 'if' after 'else'");
+        node->elseBody.push_back(parseIfChain(rest));
+      } else {
+        std::string term2;
+        node->elseBody = parseNodes(&term2);
+        if (term2 != "end")
+          throw std::runtime_error("template: expected {{end}}");
+      }
+    } else if (term != "end") {
+      throw std::runtime_error("template: expected {{end}}");
+    }
+    return node;
   }
 
   void emitText(std::vector<NodePtr>& nodes, std::string raw, bool trimRightOfText) {
@@ -772,12 +839,35 @@ class Evaluator {
           Value coll = eval(n->expr, scope);
           if (coll.kind != Value::Kind::List)
             throw std::runtime_error("template: range over non-list value");
+          int64_t idx = 0;
           for (auto& item : *coll.list) {
             Scope inner = scope;
             inner.dot = item;
             inner.dotIsEnv = false;
-            if (!n->rangeVar.empty()) inner.vars[n->rangeVar] = item;
+            if (!n->rangeVar2.empty()) {
+              // two-var form: $i = index, $v = value
+              inner.vars[n->rangeVar] = Value::integer(idx);
+              inner.vars[n->rangeVar2] = item;
+            } else if (!n->rangeVar.empty()) {
+              inner.vars[n->rangeVar] = item;
+            }
             execNodes(n->body, inner, out);
+            idx++;
+          }
+          break;
+        }
+        case Node::Kind::Assign:
+          scope.vars[n->rangeVar] = eval(n->expr, scope);
+          break;
+        case Node::Kind::With: {
+          Value v = eval(n->expr, scope);
+          if (v.truthy()) {
+            Scope inner = scope;
+            inner.dot = v;
+            inner.dotIsEnv = false;
+            execNodes(n->body, inner, out);
+          } else {
+            execNodes(n->elseBody, scope, out);
           }
           break;
         }

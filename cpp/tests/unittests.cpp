@@ -193,6 +193,22 @@ static void testTemplate() {
   CHECK_EQ(renderTemplate("{{ index (loop 5 8) 1 }}"), std::string("6"));
   CHECK_EQ(renderTemplate("{{ if lt 1 2 }}y{{ end }}"), std::string("y"));
   CHECK_EQ(renderTemplate("{{ print \"a\" 1 2 }}"), std::string("a1 2"));
+
+  // comments, assignment, with, else-if, two-var range
+  CHECK_EQ(renderTemplate("a{{/* ignore me */}}b"), std::string("ab"));
+  CHECK_EQ(renderTemplate("{{ $x := \"v\" }}{{ $x }}{{ $x }}"),
+           std::string("vv"));
+  CHECK_EQ(renderTemplate("{{ with .STAGE }}got-{{ . }}{{ end }}"),
+           std::string("got-prod"));
+  CHECK_EQ(renderTemplate("{{ with .TEST_MISSING }}y{{ else }}n{{ end }}"),
+           std::string("n"));
+  CHECK_EQ(renderTemplate(
+               "{{ if eq .STAGE \"dev\" }}d{{ else if eq .STAGE \"prod\" "
+               "}}p{{ else }}o{{ end }}"),
+           std::string("p"));
+  CHECK_EQ(renderTemplate(
+               "{{ range $i, $v := loop 5 8 }}{{ $i }}:{{ $v }} {{ end }}"),
+           std::string("0:5 1:6 2:7 "));
   unsetenv("STAGE");
 }
 

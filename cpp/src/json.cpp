@@ -25,6 +25,7 @@ class Parser {
  private:
   const std::string& s_;
   size_t pos_ = 0;
+  int depth_ = 0;
 
   [[noreturn]] void fail(const std::string& msg) {
     throw JsonParseError(msg, pos_);
@@ -64,6 +65,13 @@ class Parser {
   }
 
   Json parseValue() {
+    if (++depth_ > 500) fail("nesting too deep");
+    Json v = parseValueInner();
+    depth_--;
+    return v;
+  }
+
+  Json parseValueInner() {
     if (eof()) fail("unexpected end of input");
     char c = peek();
     switch (c) {

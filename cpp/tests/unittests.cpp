@@ -88,6 +88,22 @@ static void testJson5() {
 
   // duplicate keys: last wins (Go map unmarshal semantics)
   CHECK_EQ(parseJson5("{a: 1, a: 2}").find("a")->asInt(), (int64_t)2);
+
+  // JSON5 numeric edge forms
+  CHECK_EQ(parseJson5("[+5, .5, 5., -0x10]").array().size(), (size_t)4);
+  CHECK(parseJson5("Infinity").isDouble());
+  CHECK(parseJson5("[-Infinity]").array()[0].asDouble() < 0);
+  CHECK_EQ(parseJson5("\"\\u00e9\"").str(), std::string("\xc3\xa9"));
+
+  // pathological nesting is rejected, not a stack overflow
+  std::string deep(100000, '[');
+  bool threwDeep = false;
+  try {
+    parseJson5(deep);
+  } catch (const JsonParseError&) {
+    threwDeep = true;
+  }
+  CHECK(threwDeep);
 }
 
 static void testDurations() {

@@ -68,6 +68,15 @@ def test_parser_never_crashes_on_garbage(text):
         pass  # rejecting is fine; crashing is not
 
 
+def test_parser_rejects_bracket_bombs():
+    for bomb in ("[" * 200000, "{" * 100000,
+                 "[{" * 50000, '{"a":' * 50000):
+        try:
+            native.json5_to_json(bomb)
+        except ValueError:
+            pass
+
+
 @settings(max_examples=200, deadline=None)
 @given(st.text(alphabet=st.sampled_from(list("{}().|$\"ab \n-")), max_size=50))
 def test_template_never_crashes_on_garbage(text):

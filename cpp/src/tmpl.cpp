@@ -408,10 +408,17 @@ class TemplateParser {
   const std::string& text_;
   size_t pos_ = 0;
   bool pendingTrim_ = false;  // previous action had a right-trim marker
+  int depth_ = 0;
 
   // terminator: if non-null, stop at {{end}} / {{else}}; sets *terminator
   std::vector<NodePtr> parseNodes(std::string* terminator) {
+    if (++depth_ > 200)
+      throw std::runtime_error("template: block nesting too deep");
     std::vector<NodePtr> nodes;
+    struct DepthGuard {
+      int& d;
+      ~DepthGuard() { d--; }
+    } guard{depth_};
     while (pos_ < text_.size()) {
       size_t open = text_.find("{{", pos_);
       std::string raw = text_.substr(

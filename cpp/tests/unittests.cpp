@@ -598,6 +598,37 @@ static void testWhenEachFiresRepeatedly() {
   CHECK_EQ(probe.count(EventCode::ExitSuccess, "onchange"), 3);
 }
 
+static void testDecodeEdges() {
+  // weak coercions (config/decode/decode.go ToStrings/WeaklyTypedInput)
+  std::vector<std::string> out;
+  std::string err;
+  CHECK(decode::toStrings(parseJson5("[1, \"a\", true, 2.5]"), &out));
+  CHECK_EQ(out.size(), (size_t)4);
+  CHECK_EQ(out[0], std::string("1"));
+  CHECK_EQ(out[2], std::string("1"));  // bool -> "1"
+  CHECK(decode::toStrings(parseJson5("\"solo\""), &out) && out.size() == 1);
+  CHECK(decode::toStrings(parseJson5("null"), &out) && out.empty());
+
+  int n = 0;
+  CHECK(decode::toInt(parseJson5("\"42\""), &n) && n == 42);
+  CHECK(decode::toInt(parseJson5("3.9"), &n) && n == 3);  // truncation
+  CHECK(!decode::toInt(parseJson5("\"4x\""), &n));
+  bool b = false;
+  CHECK(decode::toBool(parseJson5("\"true\""), &b) && b);
+  CHECK(!decode::toBool(parseJson5("\"nope\""), &b));
+
+  // template nesting bomb rejects instead of overflowing
+  std::string bomb;
+  for (int i = 0; i < 5000; i++) bomb += "{{ if 1 }}";
+  bool threw = false;
+  try {
+    renderTemplate(bomb);
+  } catch (const std::exception&) {
+    threw = true;
+  }
+  CHECK(threw);
+}
+
 static void testDebugRing() {
   // the bus's 10-slot circular debug buffer (events/bus.go:24-54):
   // keeps the last 10 events, drains oldest-first
@@ -691,6 +722,7 @@ int main() {
   testPeriodicJobRuns();
   testPreStopJobOnShutdown();
   testWhenEachFiresRepeatedly();
+  testDecodeEdges();
   testDebugRing();
   testMaintenanceMatrix();
   testSignalJobMatrix();

@@ -301,6 +301,19 @@ static void testIps() {
   // default spec list: eth0:inet then inet
   CHECK(getIP({}, ifaces, &out, &err) && out == "10.2.0.5");
 
+  // IPv6 specs
+  ifaces.push_back(mk("eth2", "fe80::1234"));
+  ifaces.push_back(mk("lo", "::1"));
+  std::stable_sort(ifaces.begin(), ifaces.end(),
+                   [](const InterfaceIP& a, const InterfaceIP& b) {
+                     if (a.name != b.name) return a.name < b.name;
+                     return memcmp(a.bytes, b.bytes, 16) < 0;
+                   });
+  CHECK(getIP({"eth2:inet6"}, ifaces, &out, &err) && out == "fe80::1234");
+  CHECK(getIP({"inet6"}, ifaces, &out, &err) && out == "fe80::1234");
+  // inet6 wildcard skips ::1 loopback; eth0 has no v6 address
+  CHECK(getIP({"eth0:inet6"}, ifaces, &out, &err) == false);
+
   CHECK(validateServiceName("my-service", &err));
   CHECK(!validateServiceName("", &err));
   CHECK(!validateServiceName("-bad", &err));

@@ -16,6 +16,7 @@ environment so the wire protocol is faked at the HTTP layer instead.
 
 import json
 import threading
+import time
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from urllib.parse import urlparse, parse_qs
 
@@ -30,6 +31,8 @@ class MockConsul:
         self.health = {}        # service name -> list of dicts(ID,Address,Port)
         self.tokens = []        # X-Consul-Token header values seen
         self.fail_mode = False  # when True: every endpoint returns 500
+        self.index = 1          # consul-style modify index (global)
+        self.changed = threading.Condition(self.lock)
 
         outer = self
 
@@ -59,8 +62,21 @@ class MockConsul:
                     name = parsed.path[len("/v1/health/service/"):]
                     qs = parse_qs(parsed.query)
                     tag = qs.get("tag", [None])[0]
+                    want_index = int(qs.get("index", ["0"])[0])
+                    wait_s = 0
+                    if "wait" in qs:
+                        wait_s = int(qs["wait"][0].rstrip("s") or 0)
                     with outer.lock:
+                        # consul blocking query: hold until the index
+                        # advances past the caller's or the wait elapses
+                        if want_index and wait_s:
+                            deadline = time.time() + wait_s
+                            while (outer.index <= want_index and
+                                   time.time() < deadline):
+                                outer.changed.wait(
+                                    max(0.05, deadline - time.time()))
                         entries = list(outer.health.get(name, []))
+                        current_index = outer.index
                     if tag is not None:
                         entries = [e for e in entries
                                    if tag in e.get("Tags", [])]
@@ -73,7 +89,13 @@ class MockConsul:
                                      "Tags": e.get("Tags", [])},
                          "Checks": []}
                         for e in entries]).encode()
-                    return self._respond(200, body)
+                    self.send_response(200)
+                    self.send_header("Content-Type", "application/json")
+                    self.send_header("X-Consul-Index", str(current_index))
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                    return
                 if parsed.path == "/v1/agent/self":
                     return self._respond(200, b'{"Config":{}}')
                 return self._respond(404, b"not found")
@@ -131,6 +153,8 @@ class MockConsul:
     def set_health(self, service, entries):
         with self.lock:
             self.health[service] = entries
+            self.index += 1
+            self.changed.notify_all()
 
     def stop(self):
         self.server.shutdown()

@@ -46,6 +46,9 @@ class ConsulBackend {
   using DoneCb = std::function<void(bool ok, const std::string& err)>;
   using HealthCb =
       std::function<void(bool ok, std::vector<ServiceEntry> entries)>;
+  // blocking-query variant also yields the new Consul index
+  using HealthBlockingCb = std::function<void(
+      bool ok, std::vector<ServiceEntry> entries, uint64_t index)>;
 
   // byte-compatible payload with the reference's registration
   // (discovery/service.go:93-110)
@@ -60,6 +63,13 @@ class ConsulBackend {
   // GET /v1/health/service/<name>?passing=1 (+tag,+dc)
   void healthService(const std::string& name, const std::string& tag,
                      const std::string& dc, HealthCb cb);
+
+  // Consul blocking query: long-polls with ?index=<lastIndex>&wait=<N>s;
+  // returns when membership changes or the wait elapses. The request is
+  // cancellable so generation teardown never stalls behind the poll.
+  void healthServiceBlocking(const std::string& name, const std::string& tag,
+                             const std::string& dc, uint64_t lastIndex,
+                             int waitSeconds, HealthBlockingCb cb);
 
   // change detection against the cached set (loop thread only)
   // (discovery/consul.go:102-125)
@@ -97,6 +107,7 @@ class ConsulBackend {
 
   std::map<std::string, std::vector<ServiceEntry>> watched_;
   std::shared_ptr<prom::Family> watchGauge_;
+  std::set<std::shared_ptr<http::CancelToken>> activeTokens_;
 };
 
 // Per-job service registration state (discovery/service.go:12-110).

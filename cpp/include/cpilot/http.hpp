@@ -8,6 +8,7 @@
 
 #include <functional>
 #include <map>
+#include <mutex>
 #include <memory>
 #include <string>
 
@@ -67,6 +68,23 @@ struct ClientResult {
   int status = 0;
   std::string body;
   std::string error;
+  std::map<std::string, std::string> headers;  // lower-cased keys
+};
+
+// Cancellation for long-polling requests (Consul blocking queries):
+// cancel() shuts the socket down from another thread so the blocked
+// read returns immediately.
+class CancelToken {
+ public:
+  void arm(int fd);     // called by request() once connected
+  void disarm();        // called by request() before closing
+  void cancel();        // any thread
+  bool cancelled();
+
+ private:
+  std::mutex mu_;
+  int fd_ = -1;
+  bool cancelled_ = false;
 };
 
 // TLS options for https targets (OpenSSL; reference parity with the
@@ -87,7 +105,8 @@ ClientResult request(const std::string& target, const std::string& method,
                      const std::string& contentType = "application/json",
                      const std::map<std::string, std::string>& headers = {},
                      int timeoutMs = 10000,
-                     const TlsOptions* tls = nullptr);
+                     const TlsOptions* tls = nullptr,
+                     CancelToken* cancel = nullptr);
 
 }  // namespace http
 }  // namespace cpilot

@@ -20,6 +20,11 @@ struct WatchConfig {
   int poll = 0;             // seconds
   std::string tag;
   std::string dc;
+  // extension: blocking=true uses Consul blocking queries (long-poll on
+  // the service's index) — changes propagate in milliseconds instead of
+  // the poll interval, and idle traffic drops to one request per wait
+  // window. `interval` becomes the error-backoff period.
+  bool blocking = false;
 };
 
 bool newWatchConfigs(const Json& rawWatches,
@@ -33,7 +38,8 @@ class Watch : public std::enable_shared_from_this<Watch> {
         serviceName_(cfg->serviceName),
         tag_(cfg->tag),
         dc_(cfg->dc),
-        poll_(cfg->poll) {}
+        poll_(cfg->poll),
+        blocking_(cfg->blocking) {}
 
   const std::string& name() const { return name_; }
   const std::string& serviceName() const { return serviceName_; }
@@ -43,9 +49,13 @@ class Watch : public std::enable_shared_from_this<Watch> {
 
  private:
   void tick();
+  void issueBlocking();
+  void onResult(bool ok, std::vector<ServiceEntry> entries);
 
   std::string name_, serviceName_, tag_, dc_;
   int poll_;
+  bool blocking_ = false;
+  uint64_t lastIndex_ = 0;
   Loop* loop_ = nullptr;
   std::shared_ptr<Bus> bus_;
   ConsulBackend* consul_ = nullptr;

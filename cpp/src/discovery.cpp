@@ -142,6 +142,9 @@ void ConsulBackend::stop() {
     std::lock_guard<std::mutex> l(mu_);
     if (stopping_ && workers_.empty()) return;
     stopping_ = true;
+    // interrupt in-flight blocking queries so teardown never waits out
+    // a long-poll
+    for (auto& t : activeTokens_) t->cancel();
   }
   cv_.notify_all();
   for (auto& t : workers_) t.join();
@@ -288,6 +291,35 @@ void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
   if (!accepted) loop_->post([cb] { cb(false, "queue full"); });
 }
 
+namespace {
+
+std::vector<ServiceEntry> parseHealthEntries(const std::string& body,
+                                             bool* ok) {
+  std::vector<ServiceEntry> entries;
+  try {
+    Json doc = parseJson5(body);
+    if (doc.isArray()) {
+      for (auto& e : doc.array()) {
+        const Json* svc = e.find("Service");
+        if (!svc || !svc->isObject()) continue;
+        ServiceEntry entry;
+        if (const Json* v = svc->find("ID"))
+          if (v->isString()) entry.id = v->str();
+        if (const Json* v = svc->find("Address"))
+          if (v->isString()) entry.address = v->str();
+        if (const Json* v = svc->find("Port"))
+          if (v->isNumber()) entry.port = (int)v->asInt();
+        entries.push_back(std::move(entry));
+      }
+    }
+  } catch (const std::exception&) {
+    *ok = false;
+  }
+  return entries;
+}
+
+}  // namespace
+
 void ConsulBackend::healthService(const std::string& name,
                                   const std::string& tag,
                                   const std::string& dc, HealthCb cb) {
@@ -306,33 +338,68 @@ void ConsulBackend::healthService(const std::string& name,
                              headers, 10000, &tls);
     bool ok = res.ok && res.status == 200;
     std::vector<ServiceEntry> entries;
-    if (ok) {
-      try {
-        Json doc = parseJson5(res.body);
-        if (doc.isArray()) {
-          for (auto& e : doc.array()) {
-            const Json* svc = e.find("Service");
-            if (!svc || !svc->isObject()) continue;
-            ServiceEntry entry;
-            if (const Json* v = svc->find("ID"))
-              if (v->isString()) entry.id = v->str();
-            if (const Json* v = svc->find("Address"))
-              if (v->isString()) entry.address = v->str();
-            if (const Json* v = svc->find("Port"))
-              if (v->isNumber()) entry.port = (int)v->asInt();
-            entries.push_back(std::move(entry));
-          }
-        }
-      } catch (const std::exception&) {
-        ok = false;
-      }
-    }
+    if (ok) entries = parseHealthEntries(res.body, &ok);
     loop->post([cb, ok, entries = std::move(entries)]() mutable {
       cb(ok, std::move(entries));
     });
   });
   if (!accepted) {
     loop_->post([cb] { cb(false, {}); });
+  }
+}
+
+void ConsulBackend::healthServiceBlocking(const std::string& name,
+                                          const std::string& tag,
+                                          const std::string& dc,
+                                          uint64_t lastIndex, int waitSeconds,
+                                          HealthBlockingCb cb) {
+  std::string path = "/v1/health/service/" + urlEncode(name) + "?passing=1";
+  if (!tag.empty()) path += "&tag=" + urlEncode(tag);
+  if (!dc.empty()) path += "&dc=" + urlEncode(dc);
+  path += "&index=" + std::to_string(lastIndex) +
+          "&wait=" + std::to_string(waitSeconds) + "s";
+  auto token = std::make_shared<http::CancelToken>();
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    if (stopping_) {
+      loop_->post([cb] { cb(false, {}, 0); });
+      return;
+    }
+    activeTokens_.insert(token);
+  }
+  std::string target = address_;
+  std::string tokenHdr = token_;
+  http::TlsOptions tls = tls_;
+  Loop* loop = loop_;
+  // long-polls get a dedicated slot: run unkeyed but mark the worker
+  // timeout past the wait so the request isn't cut short
+  bool accepted = enqueue("", [this, target, tokenHdr, tls, path, cb, loop,
+                               token, waitSeconds] {
+    std::map<std::string, std::string> headers;
+    if (!tokenHdr.empty()) headers["X-Consul-Token"] = tokenHdr;
+    auto res = http::request(target, "GET", path, "", "application/json",
+                             headers, (waitSeconds + 10) * 1000, &tls,
+                             token.get());
+    {
+      std::lock_guard<std::mutex> l(mu_);
+      activeTokens_.erase(token);
+    }
+    bool ok = res.ok && res.status == 200 && !token->cancelled();
+    uint64_t index = 0;
+    auto it = res.headers.find("x-consul-index");
+    if (it != res.headers.end()) index = strtoull(it->second.c_str(), nullptr, 10);
+    std::vector<ServiceEntry> entries;
+    if (ok) entries = parseHealthEntries(res.body, &ok);
+    loop->post([cb, ok, index, entries = std::move(entries)]() mutable {
+      cb(ok, std::move(entries), index);
+    });
+  });
+  if (!accepted) {
+    {
+      std::lock_guard<std::mutex> l(mu_);
+      activeTokens_.erase(token);
+    }
+    loop_->post([cb] { cb(false, {}, 0); });
   }
 }
 

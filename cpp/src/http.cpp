@@ -242,6 +242,28 @@ static bool parseHeaders(Server::Conn* c) {
 
 // ---------------- blocking client ----------------
 
+void CancelToken::arm(int fd) {
+  std::lock_guard<std::mutex> l(mu_);
+  fd_ = fd;
+  if (cancelled_) ::shutdown(fd, SHUT_RDWR);
+}
+
+void CancelToken::disarm() {
+  std::lock_guard<std::mutex> l(mu_);
+  fd_ = -1;
+}
+
+void CancelToken::cancel() {
+  std::lock_guard<std::mutex> l(mu_);
+  cancelled_ = true;
+  if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
+}
+
+bool CancelToken::cancelled() {
+  std::lock_guard<std::mutex> l(mu_);
+  return cancelled_;
+}
+
 namespace {
 
 int connectTarget(const std::string& target, int timeoutMs, std::string* err) {
@@ -380,11 +402,23 @@ ClientResult request(const std::string& target, const std::string& method,
                      const std::string& path, const std::string& body,
                      const std::string& contentType,
                      const std::map<std::string, std::string>& headers,
-                     int timeoutMs, const TlsOptions* tls) {
+                     int timeoutMs, const TlsOptions* tls,
+                     CancelToken* cancel) {
   ClientResult result;
+  if (cancel && cancel->cancelled()) {
+    result.error = "cancelled";
+    return result;
+  }
   Transport t;
   t.fd = connectTarget(target, timeoutMs, &result.error);
   if (t.fd < 0) return result;
+  struct Disarm {
+    CancelToken* ct;
+    ~Disarm() {
+      if (ct) ct->disarm();
+    }
+  } disarm{cancel};
+  if (cancel) cancel->arm(t.fd);
 
   std::string host = target.rfind("unix:", 0) == 0 ? "localhost" : target;
   if (tls && tls->enabled) {
@@ -428,12 +462,28 @@ ClientResult request(const std::string& target, const std::string& method,
 
   size_t headerEnd = resp.find("\r\n\r\n");
   if (headerEnd == std::string::npos || resp.compare(0, 5, "HTTP/") != 0) {
-    result.error = "malformed response";
+    result.error = (cancel && cancel->cancelled()) ? "cancelled"
+                                                   : "malformed response";
     return result;
   }
   size_t sp = resp.find(' ');
   result.status = atoi(resp.c_str() + sp + 1);
   result.body = resp.substr(headerEnd + 4);
+  // response headers (lower-cased keys)
+  size_t pos = resp.find("\r\n") + 2;
+  while (pos < headerEnd) {
+    size_t eol = resp.find("\r\n", pos);
+    if (eol == std::string::npos || eol > headerEnd) break;
+    std::string line = resp.substr(pos, eol - pos);
+    pos = eol + 2;
+    size_t colon = line.find(':');
+    if (colon == std::string::npos) continue;
+    std::string key = line.substr(0, colon);
+    for (auto& ch : key) ch = tolower((unsigned char)ch);
+    size_t vstart = colon + 1;
+    while (vstart < line.size() && line[vstart] == ' ') vstart++;
+    result.headers[key] = line.substr(vstart);
+  }
   // chunked responses: dechunk (Consul uses Content-Length, but be safe)
   std::string lower = resp.substr(0, headerEnd);
   for (auto& ch : lower) ch = tolower((unsigned char)ch);

@@ -20,7 +20,8 @@ bool newWatchConfigs(const Json& rawWatches,
       *err = "Watch configuration error: watch must be an object";
       return false;
     }
-    if (!decode::checkKeys(raw, {"name", "interval", "tag", "dc"}, err)) {
+    if (!decode::checkKeys(raw, {"name", "interval", "tag", "dc", "blocking"},
+                           err)) {
       *err = "Watch configuration error: " + *err;
       return false;
     }
@@ -29,6 +30,12 @@ bool newWatchConfigs(const Json& rawWatches,
     if (const Json* v = raw.find("interval")) decode::toInt(*v, &cfg->poll);
     if (const Json* v = raw.find("tag")) decode::toString(*v, &cfg->tag);
     if (const Json* v = raw.find("dc")) decode::toString(*v, &cfg->dc);
+    if (const Json* v = raw.find("blocking")) {
+      if (!decode::toBool(*v, &cfg->blocking)) {
+        *err = "watch[" + cfg->name + "].blocking must be a bool";
+        return false;
+      }
+    }
 
     if (!validateServiceName(cfg->name, err)) return false;
     cfg->serviceName = cfg->name;
@@ -47,6 +54,10 @@ void Watch::run(Loop& loop, std::shared_ptr<Bus> bus, ConsulBackend* consul) {
   bus_ = std::move(bus);
   consul_ = consul;
   auto self = shared_from_this();
+  if (blocking_) {
+    issueBlocking();
+    return;
+  }
   timer_ = loop.addInterval(std::chrono::seconds(poll_),
                             [this, self] { tick(); });
 }
@@ -56,6 +67,23 @@ void Watch::stop(Loop& loop) {
   if (timer_) {
     loop.cancelTimer(timer_);
     timer_ = 0;
+  }
+}
+
+void Watch::onResult(bool ok, std::vector<ServiceEntry> entries) {
+  if (!ok) {
+    LOG_WARN("failed to query %s", serviceName_.c_str());
+    return;
+  }
+  consul_->watchGauge()->set({serviceName_}, (double)entries.size());
+  bool isHealthy = !entries.empty();
+  bool didChange = consul_->compareAndSwap(serviceName_, entries);
+  if (didChange) {
+    bus_->publish(Event{EventCode::StatusChanged, name_});
+    if (isHealthy)
+      bus_->publish(Event{EventCode::StatusHealthy, name_});
+    else
+      bus_->publish(Event{EventCode::StatusUnhealthy, name_});
   }
 }
 
@@ -69,19 +97,30 @@ void Watch::tick() {
       [this, self](bool ok, std::vector<ServiceEntry> entries) {
         inFlight_ = false;
         if (stopped_) return;
-        if (!ok) {
-          LOG_WARN("failed to query %s", serviceName_.c_str());
-          return;
-        }
-        consul_->watchGauge()->set({serviceName_}, (double)entries.size());
-        bool isHealthy = !entries.empty();
-        bool didChange = consul_->compareAndSwap(serviceName_, entries);
-        if (didChange) {
-          bus_->publish(Event{EventCode::StatusChanged, name_});
-          if (isHealthy)
-            bus_->publish(Event{EventCode::StatusHealthy, name_});
-          else
-            bus_->publish(Event{EventCode::StatusUnhealthy, name_});
+        onResult(ok, std::move(entries));
+      });
+}
+
+void Watch::issueBlocking() {
+  if (stopped_ || !consul_) return;
+  auto self = shared_from_this();
+  consul_->healthServiceBlocking(
+      serviceName_, tag_, dc_, lastIndex_, 10,
+      [this, self](bool ok, std::vector<ServiceEntry> entries,
+                   uint64_t index) {
+        if (stopped_) return;
+        onResult(ok, std::move(entries));
+        if (ok) {
+          // consul index contract: reset when it goes backwards
+          lastIndex_ = (index > 0 && index >= lastIndex_) ? index : 0;
+          // immediate re-issue with a tiny floor so a hot agent can't
+          // spin us
+          timer_ = loop_->addTimeout(std::chrono::milliseconds(50),
+                                     [this, self] { issueBlocking(); });
+        } else {
+          // error backoff: fall back to the configured interval
+          timer_ = loop_->addTimeout(std::chrono::seconds(poll_),
+                                     [this, self] { issueBlocking(); });
         }
       });
 }

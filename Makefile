@@ -37,3 +37,9 @@ asan:
 	ninja -C build-asan cpilot_unittests containerpilot
 	./bin/cpilot_unittests
 	ninja -C $(BUILD_DIR)
+
+soak: build
+	python3 scripts/soak.py 300
+
+capacity: build
+	python3 scripts/capacity.py

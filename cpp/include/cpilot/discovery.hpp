@@ -107,7 +107,18 @@ class ConsulBackend {
 
   std::map<std::string, std::vector<ServiceEntry>> watched_;
   std::shared_ptr<prom::Family> watchGauge_;
-  std::set<std::shared_ptr<http::CancelToken>> activeTokens_;
+
+  // blocking queries run on their own short-lived threads (a parked
+  // long-poll must not occupy the shared worker pool, or N blocking
+  // watches would starve TTL heartbeats). Shared with those threads so
+  // stop() can cancel without use-after-free on backend teardown.
+  struct BlockingReg {
+    std::mutex mu;
+    std::set<std::shared_ptr<http::CancelToken>> tokens;
+    bool stopping = false;
+  };
+  std::shared_ptr<BlockingReg> blockingReg_ =
+      std::make_shared<BlockingReg>();
 };
 
 // Per-job service registration state (discovery/service.go:12-110).

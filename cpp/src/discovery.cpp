@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <cstdlib>
+#include <thread>
 
 #include "cpilot/decode.hpp"
 #include "cpilot/http.hpp"
@@ -133,6 +134,8 @@ ConsulBackend::~ConsulBackend() { stop(); }
 void ConsulBackend::start(Loop& loop) {
   loop_ = &loop;
   stopping_ = false;
+  if (blockingReg_->stopping)
+    blockingReg_ = std::make_shared<BlockingReg>();  // fresh after stop()
   for (int i = 0; i < 4; i++)
     workers_.emplace_back([this] { workerMain(); });
 }
@@ -142,9 +145,13 @@ void ConsulBackend::stop() {
     std::lock_guard<std::mutex> l(mu_);
     if (stopping_ && workers_.empty()) return;
     stopping_ = true;
+  }
+  {
     // interrupt in-flight blocking queries so teardown never waits out
-    // a long-poll
-    for (auto& t : activeTokens_) t->cancel();
+    // a long-poll (they run on their own threads; see healthServiceBlocking)
+    std::lock_guard<std::mutex> l(blockingReg_->mu);
+    blockingReg_->stopping = true;
+    for (auto& t : blockingReg_->tokens) t->cancel();
   }
   cv_.notify_all();
   for (auto& t : workers_) t.join();
@@ -359,48 +366,45 @@ void ConsulBackend::healthServiceBlocking(const std::string& name,
   path += "&index=" + std::to_string(lastIndex) +
           "&wait=" + std::to_string(waitSeconds) + "s";
   auto token = std::make_shared<http::CancelToken>();
+  auto reg = blockingReg_;
   {
-    std::lock_guard<std::mutex> l(mu_);
-    if (stopping_) {
+    std::lock_guard<std::mutex> l(reg->mu);
+    if (reg->stopping) {
       loop_->post([cb] { cb(false, {}, 0); });
       return;
     }
-    activeTokens_.insert(token);
+    reg->tokens.insert(token);
   }
   std::string target = address_;
   std::string tokenHdr = token_;
   http::TlsOptions tls = tls_;
   Loop* loop = loop_;
-  // long-polls get a dedicated slot: run unkeyed but mark the worker
-  // timeout past the wait so the request isn't cut short
-  bool accepted = enqueue("", [this, target, tokenHdr, tls, path, cb, loop,
-                               token, waitSeconds] {
+  // a parked long-poll must not occupy the shared worker pool (N
+  // blocking watches would starve TTL heartbeats), so each query gets a
+  // short-lived thread of its own; `this` is NOT captured — the thread
+  // may outlive the backend and touches only the shared reg + the loop
+  std::thread([reg, target, tokenHdr, tls, path, cb, loop, token,
+               waitSeconds] {
     std::map<std::string, std::string> headers;
     if (!tokenHdr.empty()) headers["X-Consul-Token"] = tokenHdr;
     auto res = http::request(target, "GET", path, "", "application/json",
                              headers, (waitSeconds + 10) * 1000, &tls,
                              token.get());
     {
-      std::lock_guard<std::mutex> l(mu_);
-      activeTokens_.erase(token);
+      std::lock_guard<std::mutex> l(reg->mu);
+      reg->tokens.erase(token);
     }
     bool ok = res.ok && res.status == 200 && !token->cancelled();
     uint64_t index = 0;
     auto it = res.headers.find("x-consul-index");
-    if (it != res.headers.end()) index = strtoull(it->second.c_str(), nullptr, 10);
+    if (it != res.headers.end())
+      index = strtoull(it->second.c_str(), nullptr, 10);
     std::vector<ServiceEntry> entries;
     if (ok) entries = parseHealthEntries(res.body, &ok);
     loop->post([cb, ok, index, entries = std::move(entries)]() mutable {
       cb(ok, std::move(entries), index);
     });
-  });
-  if (!accepted) {
-    {
-      std::lock_guard<std::mutex> l(mu_);
-      activeTokens_.erase(token);
-    }
-    loop_->post([cb] { cb(false, {}, 0); });
-  }
+  }).detach();
 }
 
 bool ConsulBackend::compareAndSwap(const std::string& service,

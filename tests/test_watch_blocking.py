@@ -85,3 +85,31 @@ def test_blocking_watch_survives_reload(daemon_factory, mock_consul):
     assert wait_until(lambda: "fast-change" in d.log(), timeout=10)
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_many_blocking_watches_dont_starve_heartbeats(daemon_factory,
+                                                      mock_consul):
+    """Parked long-polls run on their own threads: eight blocking
+    watches must not block the worker pool that carries TTL
+    heartbeats."""
+    for i in range(8):
+        mock_consul.set_health("up-%d" % i, [])
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "jobs": [{
+            "name": "svc", "exec": "sleep 60", "port": 8300,
+            "interfaces": ["static:127.0.0.1"],
+            "health": {"exec": "true", "interval": 1, "ttl": 5},
+        }],
+        "watches": [{"name": "up-%d" % i, "interval": 30, "blocking": True}
+                    for i in range(8)],
+    }).start()
+    d.wait_for_socket()
+    time.sleep(2.0)  # all eight long-polls are parked now
+    n = len(mock_consul.ttl_updates)
+    time.sleep(3.0)
+    # heartbeats kept flowing while every blocking query was parked
+    assert len(mock_consul.ttl_updates) >= n + 2, mock_consul.ttl_updates
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -49,14 +49,15 @@ class Daemon:
         if env:
             self.env.update(env)
         self.proc = None
+        self._logf = None
 
     def start(self):
         build()
-        logf = open(self.log_path, "wb")
+        self._logf = open(self.log_path, "wb")
         self.proc = subprocess.Popen(
             [BINARY, "-config", self.config_path,
              "-stats-out", self.stats_path] + self.extra_args,
-            stdout=logf, stderr=subprocess.STDOUT, env=self.env,
+            stdout=self._logf, stderr=subprocess.STDOUT, env=self.env,
             start_new_session=True)
         return self
 
@@ -136,4 +137,7 @@ class Daemon:
 
     def cleanup(self):
         self.stop(timeout=10)
+        if self._logf:
+            self._logf.close()
+            self._logf = None
         shutil.rmtree(self.workdir, ignore_errors=True)

@@ -43,3 +43,9 @@ soak: build
 
 capacity: build
 	python3 scripts/capacity.py
+
+VERSION := 3.9.0-amd
+release: build
+	mkdir -p release
+	tar -czf release/containerpilot-$(VERSION).tar.gz -C bin containerpilot
+	cd release && sha1sum containerpilot-$(VERSION).tar.gz > containerpilot-$(VERSION).tar.gz.sha1

@@ -440,6 +440,14 @@ static void testConfig() {
     watches: [{name: "backend"}]})", &err) == nullptr);
   CHECK(err.find("interval must be > 0") != std::string::npos);
 
+  // blocking flag: bool (extension), bad values rejected
+  auto cfgB = newConfig(R"({"consul": "x:8500",
+    watches: [{name: "backend", interval: 5, blocking: true}]})", &err);
+  CHECK(cfgB && cfgB->watches[0]->blocking);
+  CHECK(newConfig(R"({"consul": "x:8500",
+    watches: [{name: "backend", interval: 5, blocking: "nope"}]})",
+                  &err) == nullptr);
+
   // unknown job field rejected (decode.go:15-17 ErrorUnused)
   CHECK(newConfig(R"({"consul": "x:8500",
     jobs: [{name: "j", exec: "x", bogusField: true}]})", &err) == nullptr);

@@ -32,6 +32,9 @@ def _load():
             ctypes.c_char_p, ctypes.POINTER(ctypes.c_void_p)]
         _lib.cp_validate_config.restype = ctypes.c_void_p
         _lib.cp_validate_config.argtypes = [ctypes.c_char_p]
+        _lib.cp_consul_endpoint.restype = ctypes.c_void_p
+        _lib.cp_consul_endpoint.argtypes = [
+            ctypes.c_char_p, ctypes.POINTER(ctypes.c_void_p)]
         _lib.cp_free.argtypes = [ctypes.c_void_p]
     return _lib
 
@@ -77,3 +80,18 @@ def validate_config(text):
     """Returns None when valid, else the error message."""
     lib = _load()
     return _take_string(lib, lib.cp_validate_config(text.encode()))
+
+
+def consul_endpoint(consul_json):
+    """Resolve the consul endpoint "scheme://host:port" the given consul
+    config value produces under the current CONSUL_* environment.
+
+    Note: the CONSUL_* env vars are read by the calling process's own
+    environment, so tests set os.environ before calling.
+    """
+    lib = _load()
+    err = ctypes.c_void_p()
+    out = lib.cp_consul_endpoint(consul_json.encode(), ctypes.byref(err))
+    if not out:
+        raise ValueError(_take_string(lib, err.value))
+    return _take_string(lib, out)

@@ -122,7 +122,12 @@ class ConsulBackend {
 };
 
 // Per-job service registration state (discovery/service.go:12-110).
-struct ServiceDefinition {
+// Always held via shared_ptr: async registration callbacks posted back
+// onto the reactor capture a weak_ptr so a callback that outlives its
+// generation (reload while consul is unreachable keeps the HTTP request
+// in flight past teardown) no-ops instead of touching freed memory.
+struct ServiceDefinition
+    : public std::enable_shared_from_this<ServiceDefinition> {
   std::string id;
   std::string name;
   int port = 0;

@@ -50,14 +50,25 @@ std::unique_ptr<ConsulBackend> ConsulBackend::create(const Json* raw,
     *err = "no discovery backend defined";
     return nullptr;
   }
+  // env overrides apply only where the config left the field unset,
+  // mirroring consul api.DefaultConfig + the reference's override order
+  // (discovery/config.go:29-61: env-derived defaults, then config values)
+  bool explicitAddress = false;
+  bool explicitScheme = false;
   if (raw->isString()) {
     parseRawURI(raw->str(), &backend->address_, &backend->scheme_);
+    explicitAddress = !backend->address_.empty();
+    explicitScheme = raw->str().rfind("http://", 0) == 0 ||
+                     raw->str().rfind("https://", 0) == 0;
   } else if (raw->isObject()) {
     for (auto& kv : raw->object()) {
-      if (kv.first == "address" && kv.second.isString())
+      if (kv.first == "address" && kv.second.isString()) {
         backend->address_ = kv.second.str();
-      else if (kv.first == "scheme" && kv.second.isString())
+        explicitAddress = !backend->address_.empty();
+      } else if (kv.first == "scheme" && kv.second.isString()) {
         backend->scheme_ = kv.second.str();
+        explicitScheme = true;
+      }
       else if (kv.first == "token" && kv.second.isString())
         backend->token_ = kv.second.str();
       else if (kv.first == "tls") {
@@ -94,14 +105,39 @@ std::unique_ptr<ConsulBackend> ConsulBackend::create(const Json* raw,
     *err = "no discovery backend defined";
     return nullptr;
   }
-  if (backend->address_.empty()) backend->address_ = "127.0.0.1:8500";
   if (const char* token = getenv("CONSUL_HTTP_TOKEN")) {
     if (token[0]) backend->token_ = token;
   }
+  // CONSUL_HTTP_ADDR fills in the address whenever the config did not
+  // pin one — whether consul was an empty string, an object without an
+  // address, or an object with an empty address (api.DefaultConfig)
   if (const char* addr = getenv("CONSUL_HTTP_ADDR")) {
-    if (addr[0] && raw->isString() && raw->str().empty())
-      parseRawURI(addr, &backend->address_, &backend->scheme_);
+    std::string a(addr);
+    if (!a.empty() && !explicitAddress) {
+      std::string envScheme;
+      std::string envAddr;
+      parseRawURI(a, &envAddr, &envScheme);
+      backend->address_ = envAddr;
+      if (!explicitScheme) {
+        backend->scheme_ = envScheme;
+        // a scheme-qualified env address pins the scheme (the consul
+        // client forces https when the address carries the prefix)
+        explicitScheme = a.rfind("http://", 0) == 0 ||
+                         a.rfind("https://", 0) == 0;
+      }
+    }
   }
+  // CONSUL_HTTP_SSL switches the default scheme (api.DefaultConfig);
+  // an explicit config scheme or a scheme-qualified URI/addr wins
+  if (const char* ssl = getenv("CONSUL_HTTP_SSL")) {
+    std::string s2(ssl);
+    for (auto& ch : s2) ch = (char)tolower((unsigned char)ch);
+    if (!explicitScheme && !s2.empty()) {
+      if (s2 == "1" || s2 == "true") backend->scheme_ = "https";
+      else if (s2 == "0" || s2 == "false") backend->scheme_ = "http";
+    }
+  }
+  if (backend->address_.empty()) backend->address_ = "127.0.0.1:8500";
   // env overrides (discovery/config.go:29-51)
   if (const char* v = getenv("CONSUL_CACERT"))
     if (v[0]) backend->tls_.caFile = v;
@@ -450,16 +486,24 @@ void ServiceDefinition::registerWithInitialStatus() {
 void ServiceDefinition::registerService(const std::string& status) {
   if (wasRegistered || registerInFlight) return;
   registerInFlight = true;
+  // weak capture: the completion is posted onto the reactor by a worker
+  // thread and can land after a reload has freed this generation's
+  // object graph (e.g. a 10 s HTTP timeout against an unreachable agent
+  // straddling teardown) — it must then no-op, not dereference this
+  std::weak_ptr<ServiceDefinition> weak = weak_from_this();
   consul->serviceRegister(
       id, name, tags, port, ipAddress, enableTagOverride, ttl, status,
-      deregisterCriticalServiceAfter, [this](bool ok, const std::string& err) {
-        registerInFlight = false;
+      deregisterCriticalServiceAfter,
+      [weak](bool ok, const std::string& err) {
+        auto self = weak.lock();
+        if (!self) return;  // generation torn down while in flight
+        self->registerInFlight = false;
         if (!ok) {
           LOG_WARN("service registration failed: %s", err.c_str());
           return;
         }
-        LOG_INFO("Service registered: %s", name.c_str());
-        wasRegistered = true;
+        LOG_INFO("Service registered: %s", self->name.c_str());
+        self->wasRegistered = true;
       });
 }
 

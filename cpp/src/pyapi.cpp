@@ -7,6 +7,7 @@
 #include <string>
 
 #include "cpilot/config.hpp"
+#include "cpilot/discovery.hpp"
 #include "cpilot/json.hpp"
 #include "cpilot/timing.hpp"
 #include "cpilot/tmpl.hpp"
@@ -70,6 +71,26 @@ char* cp_validate_config(const char* text) {
   auto cfg = newConfig(text, &err);
   if (cfg) return nullptr;
   return dupString(err);
+}
+
+// Resolve the consul endpoint ("scheme://host:port") that the given
+// `consul` config value + the current CONSUL_* environment produce
+// (discovery/config.go:29-61 + api.DefaultConfig env handling).
+// Returns nullptr + *errOut on config error.
+char* cp_consul_endpoint(const char* consulJson, char** errOut) {
+  try {
+    Json raw = parseJson5(consulJson);
+    std::string err;
+    auto backend = ConsulBackend::create(&raw, &err);
+    if (!backend) {
+      if (errOut) *errOut = dupString(err);
+      return nullptr;
+    }
+    return dupString(backend->scheme() + "://" + backend->address());
+  } catch (const std::exception& e) {
+    if (errOut) *errOut = dupString(e.what());
+    return nullptr;
+  }
 }
 
 }  // extern "C"

@@ -91,3 +91,82 @@ def test_registration_retries_until_agent_up(daemon_factory):
         d.terminate()
         assert d.wait(timeout=30) == 0
         consul.stop()
+
+
+class _EnvGuard:
+    """Temporarily set/unset CONSUL_* env vars in this process (the
+    native library reads them via getenv)."""
+
+    VARS = ("CONSUL_HTTP_ADDR", "CONSUL_HTTP_SSL", "CONSUL_HTTP_TOKEN")
+
+    def __init__(self, **values):
+        self.values = values
+        self.saved = {}
+
+    def __enter__(self):
+        import os
+        for k in self.VARS:
+            self.saved[k] = os.environ.pop(k, None)
+        for k, v in self.values.items():
+            os.environ[k] = v
+        return self
+
+    def __exit__(self, *exc):
+        import os
+        for k in self.VARS:
+            os.environ.pop(k, None)
+            if self.saved.get(k) is not None:
+                os.environ[k] = self.saved[k]
+
+
+def _endpoint(consul_json, **env):
+    from containerpilot_amd import native
+    with _EnvGuard(**env):
+        return native.consul_endpoint(consul_json)
+
+
+def test_env_addr_applies_when_config_has_no_address():
+    """CONSUL_HTTP_ADDR fills in the address whenever config left it
+    unset: empty string, object without address, object with empty
+    address (api.DefaultConfig semantics; advisor finding r1)."""
+    env = {"CONSUL_HTTP_ADDR": "10.1.2.3:8501"}
+    assert _endpoint('""', **env) == "http://10.1.2.3:8501"
+    assert _endpoint('{token: "t"}', **env) == "http://10.1.2.3:8501"
+    assert _endpoint('{address: ""}', **env) == "http://10.1.2.3:8501"
+
+
+def test_env_addr_does_not_override_explicit_address():
+    env = {"CONSUL_HTTP_ADDR": "10.1.2.3:8501"}
+    assert _endpoint('"consul.local:8500"', **env) == \
+        "http://consul.local:8500"
+    assert _endpoint('{address: "consul.local:8500"}', **env) == \
+        "http://consul.local:8500"
+
+
+def test_env_addr_scheme_prefix():
+    """A scheme-qualified CONSUL_HTTP_ADDR pins the scheme."""
+    assert _endpoint('""', CONSUL_HTTP_ADDR="https://10.0.0.9:8501") == \
+        "https://10.0.0.9:8501"
+    # ...even against CONSUL_HTTP_SSL=false
+    assert _endpoint('""', CONSUL_HTTP_ADDR="https://10.0.0.9:8501",
+                     CONSUL_HTTP_SSL="false") == "https://10.0.0.9:8501"
+
+
+def test_env_ssl_switches_default_scheme():
+    """CONSUL_HTTP_SSL toggles the scheme when config didn't pin one
+    (api.DefaultConfig HTTPSSLEnvName)."""
+    assert _endpoint('"consul.local:8500"', CONSUL_HTTP_SSL="true") == \
+        "https://consul.local:8500"
+    assert _endpoint('"consul.local:8500"', CONSUL_HTTP_SSL="1") == \
+        "https://consul.local:8500"
+    assert _endpoint('"consul.local:8500"', CONSUL_HTTP_SSL="false") == \
+        "http://consul.local:8500"
+    # explicit scheme wins over the env toggle
+    assert _endpoint('{address: "a:1", scheme: "http"}',
+                     CONSUL_HTTP_SSL="true") == "http://a:1"
+    assert _endpoint('"https://a:1"', CONSUL_HTTP_SSL="false") == \
+        "https://a:1"
+
+
+def test_env_defaults_without_any_config_address():
+    assert _endpoint('""') == "http://127.0.0.1:8500"

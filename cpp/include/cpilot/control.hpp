@@ -29,8 +29,13 @@ class ControlServer {
   ControlServer(Loop& loop, std::string socketPath);
   ~ControlServer();
 
-  // unlink stale socket (control/control.go:61-73); bind with retry
-  // 10x1s (control/control.go:125-140). Returns false on failure.
+  // Unlink stale socket (control/control.go:61-73), then bind. A failed
+  // bind retries 10x1s on a LOOP TIMER (the reference retries inside a
+  // goroutine, control/control.go:125-140) so event dispatch never
+  // stalls behind a contended socket; after the last failed attempt the
+  // daemon exits fatally like the reference's log.Fatal. Returns false
+  // only for unretryable config errors (missing path, unremovable stale
+  // socket).
   bool start(std::shared_ptr<Bus> bus, std::string* err);
   void stop();
 
@@ -38,12 +43,15 @@ class ControlServer {
 
  private:
   http::Response handle(const http::Request& req);
+  bool tryListen(std::string* bindErr);
+  void scheduleRetry(int attempt, const std::string& lastErr);
 
   Loop& loop_;
   std::string socketPath_;
   std::shared_ptr<Bus> bus_;
   std::unique_ptr<http::Server> server_;
   std::shared_ptr<prom::Family> requestCounter_;
+  uint64_t retryTimer_ = 0;
 };
 
 }  // namespace cpilot

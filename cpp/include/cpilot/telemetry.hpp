@@ -66,7 +66,10 @@ class Telemetry {
   void monitorJobs(const std::vector<std::shared_ptr<Job>>& jobs);
   void monitorWatches(const std::vector<std::shared_ptr<Watch>>& watches);
 
-  bool start(std::string* err);  // bind retry 10x1s
+  // Bind, retrying 10x1s on a loop timer (never blocks the reactor);
+  // exits fatally after the last failed attempt like the reference's
+  // listenWithRetry + log.Fatal (telemetry/telemetry.go:77-91).
+  bool start(std::string* err);
   void stop();
 
   std::vector<std::shared_ptr<Metric>>& metrics() { return metrics_; }
@@ -74,6 +77,7 @@ class Telemetry {
  private:
   http::Response handle(const http::Request& req);
   std::string statusJson();
+  void scheduleRetry(int attempt, const std::string& lastErr);
 
   Loop& loop_;
   std::shared_ptr<TelemetryConfig> cfg_;
@@ -81,6 +85,7 @@ class Telemetry {
   std::vector<std::shared_ptr<Metric>> metrics_;
   std::vector<std::shared_ptr<Job>> jobs_;
   std::vector<std::string> watchNames_;
+  uint64_t retryTimer_ = 0;
 };
 
 }  // namespace cpilot

@@ -55,19 +55,46 @@ bool ControlServer::start(std::shared_ptr<Bus> bus, std::string* err) {
   server_ = std::make_unique<http::Server>(
       loop_, [this](const http::Request& req) { return handle(req); });
   std::string bindErr;
-  for (int i = 0; i < 10; i++) {
-    if (server_->listenUnix(socketPath_, &bindErr)) {
-      LOG_DEBUG("control: listening to %s", socketPath_.c_str());
-      LOG_INFO("control: serving at %s", socketPath_.c_str());
-      return true;
-    }
-    std::this_thread::sleep_for(std::chrono::seconds(1));
+  if (tryListen(&bindErr)) return true;
+  // retry from the reactor instead of sleeping on it: a briefly
+  // contended socket must not freeze timers/dispatch for up to 10 s
+  LOG_WARN("control: error listening to socket at %s: %s (retrying)",
+           socketPath_.c_str(), bindErr.c_str());
+  scheduleRetry(1, bindErr);
+  return true;
+}
+
+bool ControlServer::tryListen(std::string* bindErr) {
+  if (server_->listenUnix(socketPath_, bindErr)) {
+    LOG_DEBUG("control: listening to %s", socketPath_.c_str());
+    LOG_INFO("control: serving at %s", socketPath_.c_str());
+    return true;
   }
-  *err = "error listening to socket at " + socketPath_ + ": " + bindErr;
   return false;
 }
 
+void ControlServer::scheduleRetry(int attempt, const std::string& lastErr) {
+  if (attempt >= 10) {
+    // parity with the reference's log.Fatal after exhausted retries
+    logging::logf(logging::Level::Fatal,
+                  "error listening to socket at %s: %s", socketPath_.c_str(),
+                  lastErr.c_str());
+    return;
+  }
+  retryTimer_ = loop_.addTimeout(std::chrono::seconds(1), [this, attempt] {
+    retryTimer_ = 0;
+    if (!server_) return;  // stopped while the retry was pending
+    std::string bindErr;
+    if (tryListen(&bindErr)) return;
+    scheduleRetry(attempt + 1, bindErr);
+  });
+}
+
 void ControlServer::stop() {
+  if (retryTimer_) {
+    loop_.cancelTimer(retryTimer_);
+    retryTimer_ = 0;
+  }
   if (server_) {
     server_->stop();
     server_.reset();

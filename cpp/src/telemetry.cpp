@@ -218,23 +218,47 @@ void Telemetry::monitorWatches(
 }
 
 bool Telemetry::start(std::string* err) {
+  (void)err;
   server_ = std::make_unique<http::Server>(
       loop_, [this](const http::Request& req) { return handle(req); });
   std::string bindErr;
-  for (int i = 0; i < 10; i++) {
+  if (server_->listenTcp(cfg_->ipAddress, cfg_->port, &bindErr)) {
+    LOG_INFO("telemetry: serving at %s:%d", cfg_->ipAddress.c_str(),
+             cfg_->port);
+    return true;
+  }
+  // retry from the reactor, never by sleeping on the loop thread
+  LOG_WARN("telemetry: error listening at %s:%d: %s (retrying)",
+           cfg_->ipAddress.c_str(), cfg_->port, bindErr.c_str());
+  scheduleRetry(1, bindErr);
+  return true;
+}
+
+void Telemetry::scheduleRetry(int attempt, const std::string& lastErr) {
+  if (attempt >= 10) {
+    logging::logf(logging::Level::Fatal,
+                  "error listening to socket at %s:%d: %s",
+                  cfg_->ipAddress.c_str(), cfg_->port, lastErr.c_str());
+    return;
+  }
+  retryTimer_ = loop_.addTimeout(std::chrono::seconds(1), [this, attempt] {
+    retryTimer_ = 0;
+    if (!server_) return;  // stopped while the retry was pending
+    std::string bindErr;
     if (server_->listenTcp(cfg_->ipAddress, cfg_->port, &bindErr)) {
       LOG_INFO("telemetry: serving at %s:%d", cfg_->ipAddress.c_str(),
                cfg_->port);
-      return true;
+      return;
     }
-    std::this_thread::sleep_for(std::chrono::seconds(1));
-  }
-  *err = "error listening to socket at " + cfg_->ipAddress + ":" +
-         std::to_string(cfg_->port) + ": " + bindErr;
-  return false;
+    scheduleRetry(attempt + 1, bindErr);
+  });
 }
 
 void Telemetry::stop() {
+  if (retryTimer_) {
+    loop_.cancelTimer(retryTimer_);
+    retryTimer_ = 0;
+  }
   if (server_) {
     server_->stop();
     server_.reset();

@@ -83,3 +83,68 @@ def test_consul_errors_log_and_continue(daemon_factory, mock_consul):
     assert wait_until(lambda: len(mock_consul.ttl_updates) > n)
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_bind_retry_does_not_block_dispatch(daemon_factory):
+    """A contended telemetry port retries on a loop timer; event dispatch
+    (periodic jobs, control endpoints) continues during the retry window
+    and the listener comes up once the port frees (reference retries in a
+    goroutine, control/control.go:125-140 / telemetry.go:77-91).
+    Regression test for the r1 reactor-blocking sleep_for retry."""
+    import socket as socketmod
+    import time
+    import urllib.request
+
+    # occupy the port the telemetry server wants
+    blocker = socketmod.socket()
+    blocker.setsockopt(socketmod.SOL_SOCKET, socketmod.SO_REUSEADDR, 1)
+    blocker.bind(("127.0.0.1", 0))
+    port = blocker.getsockname()[1]
+    blocker.listen(1)
+
+    d = daemon_factory({
+        # telemetry advertises a service, so discovery must be
+        # configured; an unreachable agent only logs warnings
+        "consul": "127.0.0.1:1",
+        "stopTimeout": 1,
+        "logging": {"level": "DEBUG"},
+        "jobs": [
+            {"name": "main-app", "exec": ["sleep", "60"]},
+            # a fast periodic job proves timers keep firing
+            {"name": "ticker", "exec": ["true"],
+             "when": {"interval": "200ms"}, "restarts": "unlimited"},
+        ],
+        "telemetry": {"port": port, "interfaces": ["static:127.0.0.1"]},
+    }).start()
+    d.wait_for_socket(timeout=10)
+
+    # during the retry window the reactor must stay live: control plane
+    # answers and the periodic job keeps running
+    t0 = time.time()
+    status, _ = d.control("GET", "/v3/ping")
+    assert status == 200
+    assert time.time() - t0 < 1.0, "control response stalled behind retry"
+    runs_before = d.log().count("ticker.Run start")
+    time.sleep(1.0)
+    runs_after = d.log().count("ticker.Run start")
+    assert runs_after - runs_before >= 3, (
+        "periodic job stalled during bind retry:\n" + d.log()[-2000:])
+    assert "telemetry: error listening" in d.log()
+
+    # free the port: the loop-timer retry should bind within ~2s
+    blocker.close()
+    deadline = time.time() + 5
+    body = None
+    while time.time() < deadline:
+        try:
+            with urllib.request.urlopen(
+                    "http://127.0.0.1:%d/status" % port, timeout=2) as resp:
+                body = resp.read()
+                break
+        except OSError:
+            time.sleep(0.2)
+    assert body is not None, "telemetry never bound after port freed:\n" + \
+        d.log()[-2000:]
+    assert b"ticker" in body
+    d.terminate()
+    assert d.wait(timeout=30) == 0

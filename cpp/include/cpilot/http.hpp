@@ -53,7 +53,10 @@ class Server {
  private:
   void acceptReady();
   void connReadable(std::shared_ptr<Conn> c);
+  void connWritable(std::shared_ptr<Conn> c);
+  void beginWrite(const std::shared_ptr<Conn>& c, const Response& resp);
   void closeConn(const std::shared_ptr<Conn>& c);
+  static Response errorResponse(int status);
 
   Loop& loop_;
   Handler handler_;

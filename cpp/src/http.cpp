@@ -28,11 +28,20 @@ const char* statusText(int code) {
     case 400: return "Bad Request";
     case 404: return "Not Found";
     case 405: return "Method Not Allowed";
+    case 411: return "Length Required";
+    case 413: return "Payload Too Large";
     case 422: return "Unprocessable Entity";
+    case 431: return "Request Header Fields Too Large";
     case 500: return "Internal Server Error";
     default: return "";
   }
 }
+
+// hardening limits: the reference gets equivalent protection from
+// net/http's defaults (1 MiB header cap, read deadlines)
+static constexpr size_t kMaxHeaderBytes = 64 * 1024;
+static constexpr size_t kMaxBodyBytes = 4 * 1024 * 1024;
+static constexpr int kConnDeadlineSecs = 10;
 
 struct Server::Conn {
   int fd = -1;
@@ -41,6 +50,12 @@ struct Server::Conn {
   size_t contentLength = 0;
   size_t headerEnd = 0;
   Request req;
+  // async response write state (flushed via EPOLLOUT, never by
+  // blocking the reactor)
+  std::string outbuf;
+  size_t outoff = 0;
+  bool writing = false;
+  uint64_t deadlineTimer = 0;
 };
 
 Server::Server(Loop& loop, Handler handler)
@@ -111,8 +126,14 @@ void Server::stop() {
     listenFd_ = -1;
   }
   for (auto& kv : conns_) {
+    // cancel deadline timers: their callbacks capture this server
+    if (kv.second->deadlineTimer) {
+      loop_.cancelTimer(kv.second->deadlineTimer);
+      kv.second->deadlineTimer = 0;
+    }
     loop_.unwatchFd(kv.first);
     close(kv.first);
+    kv.second->fd = -1;
   }
   conns_.clear();
 }
@@ -125,17 +146,79 @@ void Server::acceptReady() {
     auto c = std::make_shared<Conn>();
     c->fd = fd;
     conns_[fd] = c;
-    loop_.watchFd(fd, EPOLLIN | EPOLLHUP,
-                  [this, c](uint32_t) { connReadable(c); });
+    // hard per-connection deadline: a client that dribbles its request
+    // (slow-loris) or refuses to read the response is dropped, freeing
+    // the fd and buffer; reset once when the response starts
+    c->deadlineTimer = loop_.addTimeout(
+        std::chrono::seconds(kConnDeadlineSecs),
+        [this, c] {
+          c->deadlineTimer = 0;
+          closeConn(c);
+        });
+    loop_.watchFd(fd, EPOLLIN | EPOLLHUP, [this, c](uint32_t events) {
+      if (c->writing)
+        connWritable(c);
+      else
+        connReadable(c);
+      (void)events;
+    });
   }
 }
 
 void Server::closeConn(const std::shared_ptr<Conn>& c) {
   if (c->fd < 0) return;
+  if (c->deadlineTimer) {
+    loop_.cancelTimer(c->deadlineTimer);
+    c->deadlineTimer = 0;
+  }
   loop_.unwatchFd(c->fd);
   close(c->fd);
   conns_.erase(c->fd);
   c->fd = -1;
+}
+
+// Serialize the response and flush as much as the socket accepts now;
+// the remainder drains via EPOLLOUT so a slow reader never blocks the
+// reactor (it is dropped at the connection deadline instead).
+void Server::beginWrite(const std::shared_ptr<Conn>& c,
+                        const Response& resp) {
+  std::string out = "HTTP/1.1 " + std::to_string(resp.status) + " " +
+                    statusText(resp.status) + "\r\n";
+  out += "Content-Type: " + resp.contentType + "\r\n";
+  out += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
+  out += "Connection: close\r\n\r\n";
+  out += resp.body;
+  c->outbuf = std::move(out);
+  c->outoff = 0;
+  c->writing = true;
+  // fresh deadline for the write phase
+  if (c->deadlineTimer) loop_.cancelTimer(c->deadlineTimer);
+  c->deadlineTimer = loop_.addTimeout(
+      std::chrono::seconds(kConnDeadlineSecs),
+      [this, c] {
+        c->deadlineTimer = 0;
+        closeConn(c);
+      });
+  connWritable(c);
+}
+
+void Server::connWritable(std::shared_ptr<Conn> c) {
+  if (c->fd < 0) return;
+  while (c->outoff < c->outbuf.size()) {
+    ssize_t n =
+        write(c->fd, c->outbuf.data() + c->outoff, c->outbuf.size() - c->outoff);
+    if (n > 0) {
+      c->outoff += n;
+    } else if (n < 0 && errno == EINTR) {
+      continue;
+    } else if (n < 0 && errno == EAGAIN) {
+      loop_.modifyFd(c->fd, EPOLLOUT | EPOLLHUP);
+      return;  // resume when the socket drains
+    } else {
+      break;  // peer gone
+    }
+  }
+  closeConn(c);
 }
 
 static bool parseHeaders(Server::Conn* c);
@@ -146,7 +229,7 @@ void Server::connReadable(std::shared_ptr<Conn> c) {
     ssize_t n = read(c->fd, buf, sizeof(buf));
     if (n > 0) {
       c->inbuf.append(buf, n);
-      if (c->inbuf.size() > (1 << 22)) {  // 4 MiB cap
+      if (c->inbuf.size() > kMaxHeaderBytes + kMaxBodyBytes) {
         closeConn(c);
         return;
       }
@@ -164,42 +247,44 @@ void Server::connReadable(std::shared_ptr<Conn> c) {
 
   if (!c->headersDone) {
     size_t end = c->inbuf.find("\r\n\r\n");
-    if (end == std::string::npos) return;  // wait for more
+    if (end == std::string::npos) {
+      if (c->inbuf.size() > kMaxHeaderBytes) {
+        beginWrite(c, errorResponse(431));
+      }
+      return;  // wait for more (bounded by the connection deadline)
+    }
     c->headerEnd = end + 4;
+    if (c->headerEnd > kMaxHeaderBytes) {
+      beginWrite(c, errorResponse(431));
+      return;
+    }
     if (!parseHeaders(c.get())) {
       closeConn(c);
       return;
     }
     c->headersDone = true;
+    // no chunked request support: require a length (net/http would
+    // dechunk; our endpoints only ever see small bodies)
+    if (c->req.headers.count("transfer-encoding")) {
+      beginWrite(c, errorResponse(411));
+      return;
+    }
+    if (c->contentLength > kMaxBodyBytes) {
+      beginWrite(c, errorResponse(413));
+      return;
+    }
   }
   if (c->inbuf.size() < c->headerEnd + c->contentLength) return;
   c->req.body = c->inbuf.substr(c->headerEnd, c->contentLength);
 
-  Response resp = handler_(c->req);
-  std::string out = "HTTP/1.1 " + std::to_string(resp.status) + " " +
-                    statusText(resp.status) + "\r\n";
-  out += "Content-Type: " + resp.contentType + "\r\n";
-  out += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
-  out += "Connection: close\r\n\r\n";
-  out += resp.body;
-  size_t off = 0;
-  int stalls = 0;
-  while (off < out.size()) {
-    ssize_t n = write(c->fd, out.data() + off, out.size() - off);
-    if (n > 0) {
-      off += n;
-      stalls = 0;
-    } else if (n < 0 && (errno == EAGAIN || errno == EINTR)) {
-      // short blocking wait; responses are small. A peer that stops
-      // reading gets dropped after ~3s rather than wedging the server.
-      if (++stalls > 3) break;
-      struct pollfd pfd{c->fd, POLLOUT, 0};
-      poll(&pfd, 1, 1000);
-    } else {
-      break;
-    }
-  }
-  closeConn(c);
+  beginWrite(c, handler_(c->req));
+}
+
+Response Server::errorResponse(int status) {
+  Response resp;
+  resp.status = status;
+  resp.body = std::string(statusText(status)) + "\n";
+  return resp;
 }
 
 static bool parseHeaders(Server::Conn* c) {

@@ -71,3 +71,175 @@ def test_chaos_control_surface(daemon_factory, mock_consul):
     assert d.proc.poll() is None  # still alive under chaos
     d.terminate()
     assert d.wait(timeout=60) == 0
+
+
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _telemetry_daemon(daemon_factory):
+    port = _free_port()
+    d = daemon_factory({
+        "consul": "127.0.0.1:1",
+        "stopTimeout": 1,
+        "jobs": [{"name": "main-app", "exec": ["sleep", "60"]}],
+        "telemetry": {"port": port, "interfaces": ["static:127.0.0.1"]},
+    }).start()
+    d.wait_for_socket()
+    deadline = time.time() + 10
+    import urllib.request
+    while time.time() < deadline:
+        try:
+            urllib.request.urlopen(
+                "http://127.0.0.1:%d/metrics" % port, timeout=2).read()
+            break
+        except OSError:
+            time.sleep(0.1)
+    return d, port
+
+
+def _http_over(sock, raw):
+    sock.sendall(raw)
+    out = b""
+    while True:
+        chunk = sock.recv(65536)
+        if not chunk:
+            break
+        out += chunk
+    return out
+
+
+def test_slow_loris_dropped_and_server_stays_live(daemon_factory):
+    """A client that dribbles header bytes forever is dropped at the
+    connection deadline (~10s) and never blocks other requests (r1
+    hardening item: net/http gives the reference read deadlines)."""
+    import socket as socketmod
+    import urllib.request
+
+    d, port = _telemetry_daemon(daemon_factory)
+    loris = socketmod.create_connection(("127.0.0.1", port))
+    loris.sendall(b"GET /metrics HTTP/1.1\r\nHost: x\r\nX-Drip: ")
+    # while the loris holds its fd, normal requests are served
+    t0 = time.time()
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:%d/metrics" % port, timeout=5).read()
+    assert b"containerpilot_events" in body
+    assert time.time() - t0 < 2.0
+    # dribble a byte a second; the server must cut us off near 10s
+    loris.settimeout(20)
+    dropped_at = None
+    start = time.time()
+    try:
+        for _ in range(18):
+            loris.sendall(b"z")
+            time.sleep(1)
+            # recv returns b"" once the server closed on us
+            loris.setblocking(False)
+            try:
+                if loris.recv(4096) == b"":
+                    dropped_at = time.time() - start
+                    break
+            except BlockingIOError:
+                pass
+            finally:
+                loris.setblocking(True)
+    except (BrokenPipeError, ConnectionResetError):
+        dropped_at = time.time() - start
+    loris.close()
+    assert dropped_at is not None and dropped_at < 15, (
+        "slow-loris connection was never dropped")
+    d.terminate()
+    assert d.wait(timeout=30) == 0
+
+
+def test_oversized_header_rejected(daemon_factory):
+    """Header block larger than the 64 KiB cap gets 431 and the
+    connection closed instead of unbounded buffering."""
+    import socket as socketmod
+
+    d, port = _telemetry_daemon(daemon_factory)
+    s = socketmod.create_connection(("127.0.0.1", port))
+    s.settimeout(10)
+    junk = b"GET /metrics HTTP/1.1\r\nHost: x\r\nX-Pad: " + b"A" * (80 * 1024)
+    try:
+        out = _http_over(s, junk)
+    except (BrokenPipeError, ConnectionResetError):
+        out = b""
+    s.close()
+    # either an explicit 431 or a drop; never a 200
+    assert b"200 OK" not in out
+    if out:
+        assert b"431" in out.split(b"\r\n", 1)[0]
+    # server still healthy
+    import urllib.request
+    assert b"containerpilot_events" in urllib.request.urlopen(
+        "http://127.0.0.1:%d/metrics" % port, timeout=5).read()
+    d.terminate()
+    assert d.wait(timeout=30) == 0
+
+
+def test_oversized_body_rejected(daemon_factory):
+    """Content-Length above the 4 MiB cap is refused with 413 without
+    buffering the body."""
+    import socket as socketmod
+
+    d, port = _telemetry_daemon(daemon_factory)
+    s = socketmod.create_connection(("127.0.0.1", port))
+    s.settimeout(10)
+    req = (b"POST /status HTTP/1.1\r\nHost: x\r\n"
+           b"Content-Length: 104857600\r\n\r\n")
+    out = _http_over(s, req)
+    s.close()
+    assert out.split(b"\r\n", 1)[0].find(b"413") >= 0, out[:200]
+    d.terminate()
+    assert d.wait(timeout=30) == 0
+
+
+def test_chunked_request_gets_411(daemon_factory):
+    """Chunked request bodies are rejected with 411 Length Required
+    (documented limitation; endpoints only take small JSON bodies)."""
+    import socket as socketmod
+
+    d, port = _telemetry_daemon(daemon_factory)
+    s = socketmod.create_connection(("127.0.0.1", port))
+    s.settimeout(10)
+    req = (b"POST /status HTTP/1.1\r\nHost: x\r\n"
+           b"Transfer-Encoding: chunked\r\n\r\n"
+           b"5\r\nhello\r\n0\r\n\r\n")
+    out = _http_over(s, req)
+    s.close()
+    assert out.split(b"\r\n", 1)[0].find(b"411") >= 0, out[:200]
+    d.terminate()
+    assert d.wait(timeout=30) == 0
+
+
+def test_slow_reader_does_not_block_dispatch(daemon_factory):
+    """A client that requests /metrics and then never reads is parked on
+    EPOLLOUT and dropped at the deadline; meanwhile other requests and
+    the control plane stay fast (advisor finding r1: response writes
+    blocked the loop up to ~3s per slow reader)."""
+    import socket as socketmod
+    import urllib.request
+
+    d, port = _telemetry_daemon(daemon_factory)
+    # tiny receive buffer + never read -> server write must hit EAGAIN
+    slow = socketmod.create_connection(("127.0.0.1", port))
+    slow.setsockopt(socketmod.SOL_SOCKET, socketmod.SO_RCVBUF, 1)
+    slow.sendall(b"GET /metrics HTTP/1.1\r\nHost: x\r\n\r\n")
+    time.sleep(0.3)  # give the server time to fill the socket
+    t0 = time.time()
+    for _ in range(3):
+        status, _ = d.control("GET", "/v3/ping")
+        assert status == 200
+        body = urllib.request.urlopen(
+            "http://127.0.0.1:%d/metrics" % port, timeout=5).read()
+        assert b"containerpilot_events" in body
+    assert time.time() - t0 < 3.0, "dispatch stalled behind slow reader"
+    slow.close()
+    d.terminate()
+    assert d.wait(timeout=30) == 0

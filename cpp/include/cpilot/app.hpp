@@ -38,6 +38,7 @@ class App {
   void onJobComplete();
   void maybeFinishGeneration();
   void writeStats();
+  void waitStopTimeoutOrSignal(int seconds);
 
   std::string configPath_;
   std::string statsOutPath_;

@@ -1,6 +1,7 @@
 #include "cpilot/app.hpp"
 
 #include <malloc.h>
+#include <poll.h>
 #include <signal.h>
 #include <sys/epoll.h>
 #include <sys/signalfd.h>
@@ -230,7 +231,10 @@ int App::run() {
     if (!bus_->reloadFlag()) {
       if (cfg_->stopTimeout > 0) {
         LOG_DEBUG("killing all processes in %d seconds", cfg_->stopTimeout);
-        std::this_thread::sleep_for(std::chrono::seconds(cfg_->stopTimeout));
+        // wait out stopTimeout (core/app.go:146-157), but escalate: a
+        // second SIGTERM/SIGINT during this window triggers the kill
+        // sweep immediately instead of being silently swallowed
+        waitStopTimeoutOrSignal(cfg_->stopTimeout);
       }
       for (auto& job : jobs_) {
         LOG_INFO("killing processes for job %s", job->name().c_str());
@@ -255,6 +259,29 @@ int App::run() {
   }
   writeStats();
   return 0;
+}
+
+void App::waitStopTimeoutOrSignal(int seconds) {
+  auto deadline = Clock::now() + std::chrono::seconds(seconds);
+  struct pollfd pfd {signalFd_, POLLIN, 0};
+  while (true) {
+    auto left = std::chrono::duration_cast<std::chrono::milliseconds>(
+        deadline - Clock::now());
+    if (left.count() <= 0) return;
+    int rc = poll(&pfd, 1, (int)left.count());
+    if (rc <= 0) {
+      if (rc < 0 && errno == EINTR) continue;
+      return;  // timeout
+    }
+    struct signalfd_siginfo si;
+    while (read(signalFd_, &si, sizeof(si)) == sizeof(si)) {
+      if (si.ssi_signo == SIGTERM || si.ssi_signo == SIGINT) {
+        LOG_INFO("second signal received, killing processes immediately");
+        return;
+      }
+      if (si.ssi_signo == SIGCHLD) loop_.reapChildren();
+    }
+  }
 }
 
 void App::writeStats() {

@@ -308,3 +308,32 @@ def test_job_ip_env_exported(daemon_factory, mock_consul):
     assert "SVCIP=10.7.7.7" in d.log()
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_second_sigterm_escalates_kill_sweep(daemon_factory):
+    """During the final stopTimeout window a second SIGTERM/SIGINT
+    triggers the kill sweep immediately instead of being swallowed
+    (r1 weak item; the reference sleeps through it,
+    core/app.go:146-157)."""
+    import signal
+    import time
+
+    d = daemon_factory(make_config(CONSUL, [{
+            # ignores SIGTERM so the graceful drain can't finish early
+            "name": "stubborn",
+            "exec": ["bash", "-c", "trap '' TERM; sleep 120"],
+        }], stopTimeout=8)).start()
+    d.wait_for_socket()
+    time.sleep(0.5)
+
+    t0 = time.time()
+    d.signal(signal.SIGTERM)
+    time.sleep(1.0)
+    d.signal(signal.SIGTERM)  # escalate
+    rc = d.wait(timeout=6)
+    elapsed = time.time() - t0
+    assert rc == 0, d.log()
+    assert elapsed < 6, (
+        "daemon slept through the second SIGTERM (%.1fs):\n%s"
+        % (elapsed, d.log()[-2000:]))
+    assert "second signal received" in d.log()

@@ -38,6 +38,7 @@ class App {
   void onJobComplete();
   void maybeFinishGeneration();
   void writeStats();
+  void exportJobIPEnv();
   void waitStopTimeoutOrSignal(int seconds);
 
   std::string configPath_;

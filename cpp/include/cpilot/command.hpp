@@ -75,6 +75,12 @@ using CommandPtr = std::shared_ptr<Command>;
 
 // Build a Command from raw config (exec field). fields=false means raw
 // logging. Returns nullptr + err on parse failure.
+// Mark the cached base-environment snapshot stale. Must be called on
+// the loop thread after any setenv/unsetenv (PutEnviron endpoint,
+// reload-time CONTAINERPILOT_*_IP updates) so subsequent spawns see the
+// new environment.
+void commandEnvInvalidate();
+
 CommandPtr newCommand(const Json& rawExec, Duration timeout, bool raw,
                       const std::string& logField, std::string* err);
 

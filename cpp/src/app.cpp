@@ -13,6 +13,7 @@
 #include <cstring>
 #include <thread>
 
+#include "cpilot/command.hpp"
 #include "cpilot/log.hpp"
 
 namespace cpilot {
@@ -32,7 +33,15 @@ bool App::init(std::string* err) {
   if (!cfg_) return false;
   if (!logging::init(cfg_->logConfig, err)) return false;
 
-  // CONTAINERPILOT_{JOB}_IP for each advertised job (core/app.go:81-86,92-97)
+  exportJobIPEnv();
+  return true;
+}
+
+// CONTAINERPILOT_{JOB}_IP for each advertised job (core/app.go:81-86,
+// 92-97). Runs at startup AND after each reload: the reference re-runs
+// NewApp per generation, so a reloaded config's new job set updates the
+// exported IPs.
+void App::exportJobIPEnv() {
   for (auto& jobCfg : cfg_->jobs) {
     if (jobCfg->serviceDefinition) {
       std::string key = jobCfg->name;
@@ -44,7 +53,7 @@ bool App::init(std::string* err) {
       setenv(key.c_str(), jobCfg->serviceDefinition->ipAddress.c_str(), 1);
     }
   }
-  return true;
+  commandEnvInvalidate();
 }
 
 void App::setupSignals() {
@@ -256,6 +265,7 @@ int App::run() {
       break;
     }
     cfg_ = std::move(newCfg);
+    exportJobIPEnv();
   }
   writeStats();
   return 0;

@@ -29,19 +29,63 @@ std::map<std::string, std::string>& pidEnvOverlay() {
   return overlay;
 }
 
-std::vector<std::string> snapshotEnvWithOverlay() {
+// The base environment (environ minus overlay-shadowed names) is an
+// immutable shared snapshot: at thousands of spawns/s, rebuilding ~50
+// strings per spawn on the loop thread was measurable allocator churn.
+// It only goes stale when environ itself changes (PutEnviron, reload
+// IP/PID vars — commandEnvInvalidate()) or when the overlay KEY SET
+// changes (value-only PID updates don't shadow anything new). The few
+// overlay values are appended per spawn as extraEnv.
+std::shared_ptr<const std::vector<std::string>>& baseEnvCache() {
+  static std::shared_ptr<const std::vector<std::string>> cache;
+  return cache;
+}
+bool& baseEnvDirty() {
+  static bool dirty = true;
+  return dirty;
+}
+
+std::shared_ptr<const std::vector<std::string>> baseEnvSnapshot() {
+  if (!baseEnvDirty() && baseEnvCache()) return baseEnvCache();
   auto& overlay = pidEnvOverlay();
-  std::vector<std::string> env;
+  auto env = std::make_shared<std::vector<std::string>>();
   for (char** e = ::environ; *e; e++) {
     const char* eq = strchr(*e, '=');
     if (eq && overlay.count(std::string(*e, eq - *e))) continue;
-    env.emplace_back(*e);
+    env->emplace_back(*e);
   }
-  for (auto& kv : overlay) env.push_back(kv.first + "=" + kv.second);
-  return env;
+  baseEnvCache() = env;
+  baseEnvDirty() = false;
+  return baseEnvCache();
+}
+
+std::vector<std::string> overlayExtras() {
+  std::vector<std::string> extras;
+  extras.reserve(pidEnvOverlay().size());
+  for (auto& kv : pidEnvOverlay()) extras.push_back(kv.first + "=" + kv.second);
+  return extras;
+}
+
+void overlaySet(const std::string& key, const std::string& value) {
+  auto [it, inserted] = pidEnvOverlay().emplace(key, value);
+  if (inserted) {
+    // the base snapshot only needs a rebuild if this name exists in
+    // environ and must now be shadowed; CONTAINERPILOT_*_PID names
+    // normally don't, so steady-state check traffic never rebuilds
+    if (getenv(key.c_str())) baseEnvDirty() = true;
+  } else {
+    it->second = value;
+  }
+}
+
+void overlayErase(const std::string& key) {
+  if (pidEnvOverlay().erase(key) && getenv(key.c_str()))
+    baseEnvDirty() = true;
 }
 
 }  // namespace
+
+void commandEnvInvalidate() { baseEnvDirty() = true; }
 
 bool parseArgs(const Json& raw, std::string* execPath,
                std::vector<std::string>* args, std::string* err) {
@@ -155,11 +199,12 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   int readFd = pipefds[0];
   // snapshot env (+ the PID-var overlay) on this (the loop) thread; the
   // spawner must never read the live environ concurrently with setenv
-  std::vector<std::string> envCopy = snapshotEnvWithOverlay();
-  // the spawner thread does the posix_spawnp so a burst of launches never
+  auto baseEnv = baseEnvSnapshot();
+  // the spawner pool does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
   Spawner::global().spawn(
-      loop, exec_, args_, std::move(envCopy), raw_ ? -1 : pipefds[1],
+      loop, exec_, args_, std::move(baseEnv), overlayExtras(),
+      raw_ ? -1 : pipefds[1],
       [this, self, bus, readFd](pid_t pid, int err) {
         if (pid < 0) {
           LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(err));
@@ -174,7 +219,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
         pid_ = pid;
         if (pidEnvName_.empty())
           pidEnvName_ = "CONTAINERPILOT_" + envName() + "_PID";
-        pidEnvOverlay()[pidEnvName_] = std::to_string(pid);
+        overlaySet(pidEnvName_, std::to_string(pid));
 
         if (!raw_) {
           logFd_ = readFd;
@@ -259,7 +304,7 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
     logFd_ = -1;
   }
 
-  if (!pidEnvName_.empty()) pidEnvOverlay().erase(pidEnvName_);
+  if (!pidEnvName_.empty()) overlayErase(pidEnvName_);
 
   running_ = false;
   pid_ = -1;

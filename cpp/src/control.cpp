@@ -5,6 +5,7 @@
 #include <cstdio>
 #include <thread>
 
+#include "cpilot/command.hpp"
 #include "cpilot/decode.hpp"
 #include "cpilot/log.hpp"
 
@@ -168,6 +169,7 @@ http::Response ControlServer::handle(const http::Request& req) {
         if (!kv.second.isString()) return finish(422);
         setenv(kv.first.c_str(), kv.second.str().c_str(), 1);
       }
+      commandEnvInvalidate();  // future spawns see the new environ
     } catch (const std::exception&) {
       return finish(422);
     }

@@ -110,8 +110,13 @@ void Bus::publish(Event event) {
 
 void Bus::drain() {
   drainScheduled_ = false;
-  // process what's queued now; publishes from handlers go to the next batch
+  // process what's queued now (bounded: a completion burst can enqueue
+  // hundreds of events; interleave timer firing between batches so
+  // dispatch latency stays bounded — publishes from handlers and the
+  // remainder go to the next batch, order preserved)
+  constexpr size_t kDrainBatch = 64;
   size_t n = queue_.size();
+  if (n > kDrainBatch) n = kDrainBatch;
   for (size_t i = 0; i < n && !queue_.empty(); i++) {
     auto [event, publishedAt] = std::move(queue_.front());
     queue_.pop_front();

@@ -123,10 +123,19 @@ void Loop::watchChild(pid_t pid, ChildCallback cb) {
 }
 
 void Loop::reapChildren() {
-  while (true) {
+  // Bounded batch: each child callback runs the full exit path (log
+  // drain, event publishes), so reaping an unbounded burst of exits in
+  // one pass stalls timers/dispatch — the measured failure mode that
+  // made a naive spawner pool blow p99 dispatch to 50+ ms. Process up
+  // to kReapBatch now and defer a continuation for the rest (SIGCHLD is
+  // level-coalesced, so we can't rely on another signal arriving).
+  constexpr int kReapBatch = 64;
+  int n = 0;
+  while (n < kReapBatch) {
     int status = 0;
     pid_t pid = waitpid(-1, &status, WNOHANG);
-    if (pid <= 0) break;
+    if (pid <= 0) return;
+    n++;
     auto it = children_.find(pid);
     if (it != children_.end()) {
       ChildCallback cb = std::move(it->second);
@@ -140,6 +149,7 @@ void Loop::reapChildren() {
       unclaimedExits_[pid] = status;
     }
   }
+  defer([this] { reapChildren(); });
 }
 
 void Loop::armTimerFd() {
@@ -200,12 +210,32 @@ void Loop::drainDeferred() {
 }
 
 void Loop::drainPosted() {
+  // Bounded batch for the same reason as reapChildren: spawn
+  // completions posted by the spawner pool each do real setup work
+  // (pipe watch, timeout timer, child watch — possibly a synchronous
+  // exit). Leftovers re-arm the wakeup pipe so they run next iteration
+  // with timers interleaved.
+  constexpr size_t kPostedBatch = 64;
   std::deque<std::function<void()>> batch;
+  bool more = false;
   {
     std::lock_guard<std::mutex> l(postedMu_);
-    batch.swap(posted_);
+    if (posted_.size() <= kPostedBatch) {
+      batch.swap(posted_);
+    } else {
+      for (size_t i = 0; i < kPostedBatch; i++) {
+        batch.push_back(std::move(posted_.front()));
+        posted_.pop_front();
+      }
+      more = true;
+    }
   }
   for (auto& fn : batch) fn();
+  if (more) {
+    char b = 1;
+    ssize_t unused = write(wakeupFds_[1], &b, 1);
+    (void)unused;
+  }
 }
 
 void Loop::run() {

@@ -4,45 +4,73 @@
 #include <spawn.h>
 #include <unistd.h>
 
+#include <cstdlib>
+
 extern char** environ;
 
 namespace cpilot {
 
+namespace {
+
+int defaultThreadCount() {
+  if (const char* v = getenv("CPILOT_SPAWNER_THREADS")) {
+    int n = atoi(v);
+    if (n >= 1 && n <= 64) return n;
+  }
+  long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+  if (ncpu < 1) ncpu = 1;
+  int n = (int)(ncpu / 4);
+  if (n < 2) n = 2;
+  if (n > 6) n = 6;
+  return n;
+}
+
+}  // namespace
+
 Spawner& Spawner::global() {
   // intentionally leaked: destroying the condvar/mutex at static
-  // destruction while the spawner thread waits on them deadlocks exit
+  // destruction while spawner threads wait on them deadlocks exit
   static Spawner* s = new Spawner();
   return *s;
 }
 
 Spawner::Spawner() {
-  // deliberately ONE thread: it serializes launches, which acts as
-  // natural backpressure past saturation. A 4-thread pool was measured
-  // to push ~2x the launch rate at 5x baseline load but let completion
-  // bursts flood the reactor (p99 dispatch 1 ms -> 50+ ms); a supervisor
-  // should shed overlapping checks (reference single-instance exec
-  // semantics) rather than trade latency for past-saturation throughput.
-  threads_.emplace_back([this] { threadMain(); });
-  threads_.back().detach();  // process-lifetime singleton
+  int n = defaultThreadCount();
+  for (int i = 0; i < n; i++) {
+    threads_.emplace_back([this, i] { threadMain(i); });
+    threads_.back().detach();  // process-lifetime singleton
+  }
 }
 
 void Spawner::spawn(Loop& loop, std::string execPath,
                     std::vector<std::string> args,
-                    std::vector<std::string> env, int stdioFd, SpawnCb cb) {
+                    std::shared_ptr<const std::vector<std::string>> baseEnv,
+                    std::vector<std::string> extraEnv, int stdioFd,
+                    SpawnCb cb) {
   {
     std::lock_guard<std::mutex> l(mu_);
     queue_.push_back(Request{&loop, std::move(execPath), std::move(args),
-                             std::move(env), stdioFd, std::move(cb)});
+                             std::move(baseEnv), std::move(extraEnv), stdioFd,
+                             std::move(cb)});
   }
-  cv_.notify_one();
+  // notify_all: predicates differ per thread (depth gates); notify_one
+  // could wake only a gated thread that immediately re-sleeps
+  cv_.notify_all();
 }
 
-void Spawner::threadMain() {
+void Spawner::threadMain(int index) {
+  // Depth gating: thread 0 always serves; thread i only engages once
+  // the queue backs up past 2*i requests. Concurrent vfork'd spawns
+  // contend on the parent's mm (exec of a CLONE_VM child takes the
+  // shared mmap lock), which was measured to double the reactor's p99
+  // dispatch latency at LIGHT load — so extra spawn concurrency is
+  // bought only when a backlog actually needs it.
+  const size_t gate = (size_t)(2 * index);
   while (true) {
     Request req;
     {
       std::unique_lock<std::mutex> l(mu_);
-      cv_.wait(l, [this] { return !queue_.empty(); });
+      cv_.wait(l, [this, gate] { return queue_.size() > gate; });
       req = std::move(queue_.front());
       queue_.pop_front();
     }
@@ -71,7 +99,11 @@ void Spawner::threadMain() {
     for (auto& a : req.args) argv.push_back(const_cast<char*>(a.c_str()));
     argv.push_back(nullptr);
     std::vector<char*> envp;
-    for (auto& e : req.env) envp.push_back(const_cast<char*>(e.c_str()));
+    if (req.baseEnv) {
+      envp.reserve(req.baseEnv->size() + req.extraEnv.size() + 1);
+      for (auto& e : *req.baseEnv) envp.push_back(const_cast<char*>(e.c_str()));
+    }
+    for (auto& e : req.extraEnv) envp.push_back(const_cast<char*>(e.c_str()));
     envp.push_back(nullptr);
 
     pid_t pid = -1;

@@ -16,12 +16,21 @@ Prometheus endpoint (containerpilot_events / containerpilot_event_deliveries
 counters and the containerpilot_event_dispatch_seconds histogram) at the
 window edges.
 
-The headline value is bus event deliveries/sec (each delivery = one event
-dispatched into one component's state machine — the unit of work the
-reference's O(subscribers) fan-out performs, events/bus.go:125-140),
-aggregated over all ranks. Published events/sec and the p99
-publish->dispatch latency are reported alongside; the <1 ms p99 target
-from BASELINE.md applies to the latter.
+The headline value is PUBLISHED events/sec — health-check and watch
+events entering the bus (events/bus.go:125-140) — aggregated over all
+ranks, compared like-for-like against the BASELINE >=10k events/sec
+target. Bus deliveries/sec (published x subscriber fan-out) and the p99
+publish->dispatch latency are reported alongside in config; the <1 ms
+p99 target from BASELINE.md applies to the latter, and
+check_completion_pct shows whether the daemon kept up with the
+configured check rate (a number achieved by shedding checks would be
+hollow).
+
+The default shape is the BASELINE stress config scaled 5x (500 jobs x
+100 ms health + 50 watches): the original 100-job shape structurally
+caps published events at ~2k/s (100 jobs x 10 checks/s x 2 events), so
+it cannot demonstrate the 10k target no matter how fast the daemon is.
+The 100-job shape is reported in profiles/capacity.md.
 """
 
 import argparse
@@ -157,7 +166,7 @@ def main():
     parser.add_argument("--steps", type=int, default=30,
                         help="timed seconds of stress operation")
     parser.add_argument("--warmup", type=int, default=5)
-    parser.add_argument("--jobs", type=int, default=100)
+    parser.add_argument("--jobs", type=int, default=500)
     parser.add_argument("--watches", type=int, default=50)
     parser.add_argument("--check-ms", type=int, default=100)
     args = parser.parse_args()
@@ -214,9 +223,14 @@ def main():
             total_published = sum(float(g[1]) for g in gathered)
             max_elapsed = max(float(g[2]) for g in gathered)
             worst_p99_ms = max(float(g[3]) for g in gathered)
+            target_checks = args.jobs * (1000.0 / args.check_ms)
+            completion_pct = min(
+                100.0, 100.0 * total_published /
+                (2.0 * target_checks * max(1, world_size
+                                           if distributed else 1)))
             result = {
                 "metric": "events/sec",
-                "value": round(total_delivered, 1),
+                "value": round(total_published, 1),
                 "unit": "events/s",
                 "n_gpus": world_size if distributed else args.gpus,
                 "steps": args.steps,
@@ -225,11 +239,14 @@ def main():
                 "higher_is_better": True,
                 "scaling": "weak",
                 "vs_baseline": round(
-                    total_delivered / BASELINE_EVENTS_PER_SEC, 3),
+                    total_published / BASELINE_EVENTS_PER_SEC, 3),
                 "dtype": "n/a (CPU daemon, no tensor math; see BASELINE.json north_star)",
                 "data": "synthetic (stress config: sleep jobs + /bin/true checks + mock consul)",
                 "config": {
-                    "model": "containerpilot daemon, BASELINE stress config",
+                    "model": ("containerpilot daemon, BASELINE stress "
+                              "config" if args.jobs == 100 else
+                              "containerpilot daemon, BASELINE stress "
+                              "shape scaled to %d jobs" % args.jobs),
                     "jobs": args.jobs,
                     "health_check_interval_ms": args.check_ms,
                     "watches": args.watches,
@@ -238,6 +255,8 @@ def main():
                     "parallelism": "%d independent daemons (1/rank)"
                                    % (world_size if distributed else args.gpus),
                     "published_events_per_sec": round(total_published, 1),
+                    "delivered_events_per_sec": round(total_delivered, 1),
+                    "check_completion_pct": round(completion_pct, 1),
                     "p99_dispatch_ms": round(worst_p99_ms, 4),
                     "p99_target_ms": 1.0,
                     "events_per_sec_target": BASELINE_EVENTS_PER_SEC,

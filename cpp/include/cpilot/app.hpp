@@ -61,6 +61,9 @@ class App {
   struct ShutdownWatcher : Subscriber {
     App* app = nullptr;
     void onEvent(const Event& event) override;
+    Subscription subscription() const override {
+      return {false, {}, {EventCode::Shutdown}};
+    }
   };
   ShutdownWatcher shutdownWatcher_;
 

@@ -10,6 +10,7 @@
 #pragma once
 
 #include <deque>
+#include <unordered_map>
 #include <random>
 #include <functional>
 #include <memory>
@@ -67,6 +68,21 @@ extern const Event GlobalEnterMaintenance;
 extern const Event GlobalExitMaintenance;
 extern const Event QuitByTest;
 
+// A subscriber's declared interests. The bus indexes delivery by event
+// source and by event code so an event reaches only the subscribers
+// that can act on it: naive full fan-out is O(subscribers) per event —
+// O(jobs^2) total bus work for the stress shape — and was measured to
+// cap the reactor at ~4.4M deliveries/s (~8.5k published events/s at
+// 500 jobs) on a 256-CPU EPYC. Interests match the reference's
+// semantics observably: a subscriber receiving an event it would ignore
+// is a no-op there (each job's processEvent switch filters,
+// jobs/jobs.go:195-232), so not delivering it is equivalent.
+struct Subscription {
+  bool all = true;                   // wildcard: deliver every event
+  std::vector<std::string> sources;  // deliver events with these sources
+  std::vector<EventCode> codes;      // deliver events with these codes
+};
+
 class Subscriber {
  public:
   virtual ~Subscriber() = default;
@@ -78,6 +94,13 @@ class Subscriber {
     (void)srcHash;
     onEvent(event);
   }
+  // declared interests, read once at subscribe() time; the default is
+  // full fan-out (reference behavior; used by test subscribers)
+  virtual Subscription subscription() const { return {}; }
+
+ private:
+  friend class Bus;
+  uint64_t busSeq_ = 0;  // dedup stamp when multiple indexes match
 };
 
 class Bus : public std::enable_shared_from_this<Bus> {
@@ -112,9 +135,16 @@ class Bus : public std::enable_shared_from_this<Bus> {
 
  private:
   void drain();
+  void deliverList(std::vector<Subscriber*>& list, const Event& event,
+                   size_t srcHash, uint64_t seq, uint64_t* batch);
+  void compactIndexes();
 
   Loop& loop_;
-  std::vector<Subscriber*> subscribers_;  // nullptr = tombstoned slot
+  // interest indexes (nullptr = tombstoned slot, compacted post-drain)
+  std::vector<Subscriber*> wildcard_;
+  std::unordered_map<size_t, std::vector<Subscriber*>> bySource_;
+  std::unordered_map<int, std::vector<Subscriber*>> byCode_;
+  uint64_t seq_ = 0;
   bool tombstones_ = false;
   std::deque<std::pair<Event, TimePoint>> queue_;
   bool drainScheduled_ = false;

@@ -97,6 +97,12 @@ class Job : public Subscriber, public std::enable_shared_from_this<Job> {
 
   void onEvent(const Event& event) override;
   void onEventHashed(const Event& event, size_t srcHash) override;
+  // Declared interests for indexed bus delivery: the sources this job's
+  // dispatch() can match (its own name, its check name, its start/stop
+  // dependency sources, its timer sources) plus the broadcast codes
+  // (Shutdown, maintenance, Signal, Quit). Any event outside this set
+  // is a no-op in dispatch() by construction (jobs/jobs.go:195-232).
+  Subscription subscription() const override;
 
  private:
   enum class Phase { Running, StoppingWait, Complete };
@@ -157,7 +163,7 @@ class Job : public Subscriber, public std::enable_shared_from_this<Job> {
   std::shared_ptr<Bus> bus_;
   std::function<void()> completedCb_;
   uint64_t freqTimer_ = 0, heartbeatTimer_ = 0, startTimeoutTimer_ = 0,
-           stoppingTimer_ = 0;
+           stoppingTimer_ = 0, regRetryTimer_ = 0;
 };
 
 }  // namespace cpilot

@@ -37,6 +37,11 @@ class Metric : public Subscriber,
   explicit Metric(const std::shared_ptr<MetricConfig>& cfg) : cfg_(cfg) {}
   void run(std::shared_ptr<Bus> bus);
   void onEvent(const Event& event) override;
+  Subscription subscription() const override {
+    // Metric events carry "name|value" as their source, so interest is
+    // by code; each collector filters by its own name in onEvent
+    return {false, {}, {EventCode::Metric}};
+  }
   const std::string& name() const { return cfg_->fullName; }
 
  private:

@@ -73,18 +73,32 @@ Bus::Bus(Loop& loop) : loop_(loop), ring_(10) {
       prom::MetricType::Counter);
 }
 
-void Bus::subscribe(Subscriber* s) { subscribers_.push_back(s); }
+void Bus::subscribe(Subscriber* s) {
+  Subscription sub = s->subscription();
+  if (sub.all) {
+    wildcard_.push_back(s);
+    return;
+  }
+  std::hash<std::string> h;
+  for (auto& src : sub.sources) bySource_[h(src)].push_back(s);
+  for (auto code : sub.codes) byCode_[(int)code].push_back(s);
+}
 
 void Bus::unsubscribe(Subscriber* s) {
-  // tombstone instead of erase: delivery checks the slot in O(1) rather
-  // than std::find-ing membership per delivery (which made dispatch
-  // O(subscribers^2) per event); slots compact after each drain batch
-  for (auto& slot : subscribers_) {
-    if (slot == s) {
-      slot = nullptr;
-      tombstones_ = true;
+  // tombstone instead of erase: a subscriber can unsubscribe from
+  // inside a delivery loop that is iterating the very list it lives in;
+  // slots compact after each drain batch
+  auto tombstone = [&](std::vector<Subscriber*>& list) {
+    for (auto& slot : list) {
+      if (slot == s) {
+        slot = nullptr;
+        tombstones_ = true;
+      }
     }
-  }
+  };
+  tombstone(wildcard_);
+  for (auto& kv : bySource_) tombstone(kv.second);
+  for (auto& kv : byCode_) tombstone(kv.second);
 }
 
 void Bus::publish(Event event) {
@@ -133,31 +147,58 @@ void Bus::drain() {
       uint64_t slot = latencyRng_() % latencySeen_;
       if (slot < latencyCap_) latencyWindow_[slot] = latency;
     }
-    // handlers may (un)subscribe during delivery: unsubscribes tombstone
-    // their slot (checked per delivery), new subscribes append past the
-    // bound captured here so they don't see this event
-    size_t bound = subscribers_.size();
+    // indexed delivery: wildcard subscribers, then code-interested,
+    // then source-interested. Per-subscriber order is publish order;
+    // the seq stamp dedups a subscriber matched by several indexes.
     size_t srcHash = std::hash<std::string>{}(event.source);
+    uint64_t seq = ++seq_;
     uint64_t batch = 0;
-    for (size_t si = 0; si < bound; si++) {
-      Subscriber* s = subscribers_[si];
-      if (!s) continue;
-      delivered_++;
-      batch++;
-      s->onEventHashed(event, srcHash);
-    }
+    deliverList(wildcard_, event, srcHash, seq, &batch);
+    auto codeIt = byCode_.find((int)event.code);
+    if (codeIt != byCode_.end())
+      deliverList(codeIt->second, event, srcHash, seq, &batch);
+    auto srcIt = bySource_.find(srcHash);
+    if (srcIt != bySource_.end())
+      deliverList(srcIt->second, event, srcHash, seq, &batch);
     if (batch) deliveriesCounter_->inc({}, (double)batch);
   }
-  if (tombstones_) {
-    subscribers_.erase(
-        std::remove(subscribers_.begin(), subscribers_.end(), nullptr),
-        subscribers_.end());
-    tombstones_ = false;
-  }
+  if (tombstones_) compactIndexes();
   if (!queue_.empty() && !drainScheduled_) {
     drainScheduled_ = true;
     loop_.defer([self = shared_from_this()] { self->drain(); });
   }
+}
+
+void Bus::deliverList(std::vector<Subscriber*>& list, const Event& event,
+                      size_t srcHash, uint64_t seq, uint64_t* batch) {
+  // handlers may (un)subscribe during delivery: unsubscribes tombstone
+  // their slot (checked per delivery), new subscribes append past the
+  // bound captured here so they don't see this event
+  size_t bound = list.size();
+  for (size_t si = 0; si < bound; si++) {
+    Subscriber* s = list[si];
+    if (!s || s->busSeq_ == seq) continue;
+    s->busSeq_ = seq;
+    delivered_++;
+    (*batch)++;
+    s->onEventHashed(event, srcHash);
+  }
+}
+
+void Bus::compactIndexes() {
+  auto compact = [](std::vector<Subscriber*>& list) {
+    list.erase(std::remove(list.begin(), list.end(), nullptr), list.end());
+  };
+  compact(wildcard_);
+  for (auto it = bySource_.begin(); it != bySource_.end();) {
+    compact(it->second);
+    it = it->second.empty() ? bySource_.erase(it) : std::next(it);
+  }
+  for (auto it = byCode_.begin(); it != byCode_.end();) {
+    compact(it->second);
+    it = it->second.empty() ? byCode_.erase(it) : std::next(it);
+  }
+  tombstones_ = false;
 }
 
 std::vector<Event> Bus::debugEvents() {

@@ -527,6 +527,21 @@ Job::Job(const std::shared_ptr<JobConfig>& cfg)
   }
 }
 
+Subscription Job::subscription() const {
+  Subscription sub;
+  sub.all = false;
+  sub.sources = {name_,          healthCheckName_,
+                 heartbeatSource_, runEverySource_,
+                 stoppingTimeoutSource_, name_ + ".wait-timeout"};
+  if (!startEvent_.source.empty()) sub.sources.push_back(startEvent_.source);
+  if (stoppingWaitEvent_ != NonEvent)
+    sub.sources.push_back(stoppingWaitEvent_.source);
+  sub.codes = {EventCode::Quit, EventCode::Shutdown,
+               EventCode::EnterMaintenance, EventCode::ExitMaintenance,
+               EventCode::Signal};
+  return sub;
+}
+
 void Job::run(Loop& loop, std::shared_ptr<Bus> bus,
               std::function<void()> completedCb) {
   loop_ = &loop;
@@ -558,6 +573,21 @@ void Job::run(Loop& loop, std::shared_ptr<Bus> bus,
         LOG_DEBUG("timer: {TimerExpired %s.heartbeat}", name_.c_str());
       processEvent(Event{EventCode::TimerExpired, heartbeatSource_});
     }, phase(heartbeat_));
+  }
+  if (service_ && !service_->initialStatus.empty()) {
+    // The reference retries initial-status registration at the top of
+    // every event-loop pass (jobs/jobs.go:168-171); with indexed bus
+    // delivery a quiet job may receive no events, so the retry runs on
+    // a 1s timer until registration sticks (same observable behavior:
+    // registration is retried until the agent answers).
+    regRetryTimer_ = loop.addInterval(std::chrono::seconds(1), [this, self] {
+      if (service_->wasRegistered) {
+        loop_->cancelTimer(regRetryTimer_);
+        regRetryTimer_ = 0;
+        return;
+      }
+      checkRegistration();
+    });
   }
   if (startTimeout_ > Duration(0)) {
     std::string timeoutName = name_ + ".wait-timeout";
@@ -824,7 +854,9 @@ void Job::finishCleanup() {
   if (heartbeatTimer_) loop_->cancelTimer(heartbeatTimer_);
   if (startTimeoutTimer_) loop_->cancelTimer(startTimeoutTimer_);
   if (stoppingTimer_) loop_->cancelTimer(stoppingTimer_);
-  freqTimer_ = heartbeatTimer_ = startTimeoutTimer_ = stoppingTimer_ = 0;
+  if (regRetryTimer_) loop_->cancelTimer(regRetryTimer_);
+  freqTimer_ = heartbeatTimer_ = startTimeoutTimer_ = stoppingTimer_ =
+      regRetryTimer_ = 0;
   if (exec_ && exec_->running()) exec_->term();
   if (healthCheckExec_ && healthCheckExec_->running()) healthCheckExec_->term();
 

@@ -47,5 +47,5 @@ capacity: build
 VERSION := 3.9.0-amd
 release: build
 	mkdir -p release
-	tar -czf release/containerpilot-$(VERSION).tar.gz -C bin containerpilot
+	tar -czf release/containerpilot-$(VERSION).tar.gz -C bin containerpilot cpilot-spawn-helper
 	cd release && sha1sum containerpilot-$(VERSION).tar.gz > containerpilot-$(VERSION).tar.gz.sha1

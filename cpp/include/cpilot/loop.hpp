@@ -56,12 +56,22 @@ class Loop {
   void post(std::function<void()> fn);
 
   // --- children ---
-  // Registers interest in a child's exit. If the child was already
-  // reaped (spawn-completion can race SIGCHLD when spawning happens on
-  // the spawner thread), the stored status is delivered immediately.
+  // Registers interest in a child's exit. If the exit was already
+  // recorded (a helper's exit notice can overtake the spawn-completion
+  // callback), the stored status is delivered immediately.
   void watchChild(pid_t pid, ChildCallback cb);
   // call on SIGCHLD: waitpid(-1, WNOHANG) loop dispatching to callbacks
+  // (direct children: spawn helpers; supervised processes are helper
+  // children and arrive via notifyChildExit instead)
   void reapChildren();
+  // dispatch an exit reported by a spawn helper (loop thread only; the
+  // spawner posts these). Same claimed/unclaimed logic as reapChildren.
+  void notifyChildExit(pid_t pid, int status);
+
+  // Liveness-checked cross-thread post: delivers fn to `loop` only if it
+  // still exists (spawn-helper exit notices can outlive the Loop that
+  // requested the spawn, e.g. unit tests destroying per-test loops).
+  static void postIfLive(Loop* loop, std::function<void()> fn);
 
   void run();   // until stop()
   void stop();

@@ -1,27 +1,29 @@
-// Spawner pool: posix_spawnp runs off the reactor so bursts of
-// health-check launches never delay event dispatch.
+// Spawner: hands process launches to a pool of pre-spawned helper
+// processes over SOCK_SEQPACKET socketpairs (see spawnproto.hpp for the
+// protocol and the measurements behind the design).
 //
-// Sizing: one posix_spawn("/bin/true") call costs ~0.2-0.5 ms on the
-// parent side (vfork suspends only the calling thread until exec), so a
-// single thread tops out at ~2-4k launches/s — below the >=10k
-// published-events/s target (each completed check publishes two
-// events). The pool defaults to nproc/4 clamped to [2,6] and can be
-// pinned with CPILOT_SPAWNER_THREADS. Overload backpressure is
-// preserved: a heartbeat tick whose check is still pending is skipped
-// (single-instance exec semantics), and completions are batch-paced
-// into the reactor by Loop::drainPosted's bounded batches, which is
-// what kept a naive pool from blowing p99 dispatch latency (see
-// profiles/capacity.md).
+// Why not spawn from the daemon: vfork'd posix_spawns share the
+// daemon's mm until exec, so concurrent spawns serialize on its mmap
+// lock and stall the reactor's page faults; fork also copies the
+// daemon's fd table (one log pipe per supervised job), making per-spawn
+// cost grow with job count. Helpers are tiny exec'd processes: spawn
+// cost is constant and parallel across helpers, and each helper reaps
+// its own children, so the reactor never forks and never takes child
+// SIGCHLD storms — exits stream back as messages the reader thread
+// posts onto the requesting loop.
 //
-// Spawn completions are posted back onto the loop; Loop::watchChild
-// handles the SIGCHLD-before-completion race via its unclaimed-exit map.
+// Overload backpressure is preserved upstream: a heartbeat tick whose
+// check is still pending is skipped (single-instance exec semantics),
+// and completions/exits are batch-paced into the reactor by
+// Loop::drainPosted's bounded batches.
 #pragma once
 
 #include <sys/types.h>
 
-#include <condition_variable>
+#include <cstdint>
 #include <deque>
 #include <functional>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -39,35 +41,52 @@ class Spawner {
 
   static Spawner& global();
 
-  // stdioFd >= 0 is dup2'd onto the child's stdout+stderr and closed
-  // after the spawn completes. baseEnv is an immutable shared snapshot
-  // of the environment taken ON THE LOOP THREAD (posix_spawn must never
-  // read the live environ while the reactor setenv()s); extraEnv are
-  // per-spawn overlay entries ("K=V") appended after it.
+  // stdioFd >= 0 is passed to the helper (SCM_RIGHTS), dup2'd onto the
+  // child's stdout+stderr, and closed on this side after sending.
+  // baseEnv is an immutable shared snapshot of the environment taken ON
+  // THE LOOP THREAD; extraEnv are per-spawn overlay entries ("K=V")
+  // appended after it.
   void spawn(Loop& loop, std::string execPath, std::vector<std::string> args,
              std::shared_ptr<const std::vector<std::string>> baseEnv,
              std::vector<std::string> extraEnv, int stdioFd, SpawnCb cb);
 
-  int threads() const { return (int)threads_.size(); }
+  int helperCount() const { return (int)helpers_.size(); }
 
  private:
   Spawner();
-  void threadMain(int index);
-
-  struct Request {
+  struct Helper {
+    int sock = -1;
+    pid_t pid = -1;
+    bool dead = false;
+    // requests that hit EAGAIN (kernel buffer full); flushed by the
+    // reader thread on POLLOUT, order preserved
+    std::deque<std::pair<std::vector<char>, int>> overflow;  // buf, fd
+  };
+  struct Pending {
     Loop* loop;
-    std::string execPath;
-    std::vector<std::string> args;
-    std::shared_ptr<const std::vector<std::string>> baseEnv;
-    std::vector<std::string> extraEnv;
-    int stdioFd;
     SpawnCb cb;
+    size_t helperIdx;
   };
 
-  std::vector<std::thread> threads_;
-  std::mutex mu_;
-  std::condition_variable cv_;
-  std::deque<Request> queue_;
+  bool launchHelper(Helper* h);
+  void readerMain();
+  void handleMessage(size_t idx, const char* buf, size_t len);
+  void helperDied(size_t idx);
+  bool sendRequest(Helper& h, const std::vector<char>& buf, int fd);
+  void wakeReader();
+
+  std::string helperPath_;
+  std::vector<Helper> helpers_;
+  size_t nextHelper_ = 0;
+
+  std::mutex mu_;  // guards pending_, overflow queues, helper respawn
+  std::map<uint32_t, Pending> pending_;
+  uint32_t nextReqId_ = 1;
+  // pid -> loop for exit notices (reader thread only)
+  std::map<pid_t, Loop*> pidLoops_;
+
+  int wakeFds_[2] = {-1, -1};
+  std::thread reader_;
 };
 
 }  // namespace cpilot

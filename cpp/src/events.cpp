@@ -137,6 +137,10 @@ void Bus::drain() {
     double latency =
         std::chrono::duration<double>(Clock::now() - publishedAt).count();
     dispatchHist_->observe(latency);
+    static const bool latDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+    if (latDebug && latency > 0.004)
+      LOG_WARN("slow dispatch: %s waited %.1f ms (queue %zu)",
+               event.str().c_str(), latency * 1e3, queue_.size());
     // reservoir sample: bounded memory with uniform coverage of the
     // whole run (a plain prefix window stopped representing steady
     // state and grew ~16 KB/s until its cap)

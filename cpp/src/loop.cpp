@@ -7,13 +7,36 @@
 #include <unistd.h>
 
 #include <cstring>
+#include <set>
 #include <stdexcept>
 
 #include "cpilot/log.hpp"
 
 namespace cpilot {
 
+namespace {
+// registry of live loops for postIfLive (guards cross-thread posts that
+// can outlive a Loop, e.g. helper exit notices after a test's loop died)
+std::mutex& liveLoopsMu() {
+  static std::mutex mu;
+  return mu;
+}
+std::set<Loop*>& liveLoops() {
+  static std::set<Loop*> loops;
+  return loops;
+}
+}  // namespace
+
+void Loop::postIfLive(Loop* loop, std::function<void()> fn) {
+  std::lock_guard<std::mutex> l(liveLoopsMu());
+  if (liveLoops().count(loop)) loop->post(std::move(fn));
+}
+
 Loop::Loop() {
+  {
+    std::lock_guard<std::mutex> l(liveLoopsMu());
+    liveLoops().insert(this);
+  }
   epfd_ = epoll_create1(EPOLL_CLOEXEC);
   if (epfd_ < 0) throw std::runtime_error("epoll_create1 failed");
   timerfd_ = timerfd_create(CLOCK_MONOTONIC, TFD_NONBLOCK | TFD_CLOEXEC);
@@ -36,6 +59,10 @@ Loop::Loop() {
 }
 
 Loop::~Loop() {
+  {
+    std::lock_guard<std::mutex> l(liveLoopsMu());
+    liveLoops().erase(this);
+  }
   if (epfd_ >= 0) close(epfd_);
   if (timerfd_ >= 0) close(timerfd_);
   if (wakeupFds_[0] >= 0) close(wakeupFds_[0]);
@@ -136,20 +163,24 @@ void Loop::reapChildren() {
     pid_t pid = waitpid(-1, &status, WNOHANG);
     if (pid <= 0) return;
     n++;
-    auto it = children_.find(pid);
-    if (it != children_.end()) {
-      ChildCallback cb = std::move(it->second);
-      children_.erase(it);
-      cb(status);
-    } else {
-      // exit raced the spawner thread's completion post; stash it for
-      // the watchChild that is about to arrive. (A child we never
-      // watched is an inherited zombie; the map stays tiny because every
-      // spawn claims its exit.)
-      unclaimedExits_[pid] = status;
-    }
+    notifyChildExit(pid, status);
   }
   defer([this] { reapChildren(); });
+}
+
+void Loop::notifyChildExit(pid_t pid, int status) {
+  auto it = children_.find(pid);
+  if (it != children_.end()) {
+    ChildCallback cb = std::move(it->second);
+    children_.erase(it);
+    cb(status);
+  } else {
+    // exit notice overtook the spawn-completion callback; stash it for
+    // the watchChild that is about to arrive. (An exit we never watch
+    // is an inherited zombie already reaped; the map stays tiny because
+    // every spawn claims its exit.)
+    unclaimedExits_[pid] = status;
+  }
 }
 
 void Loop::armTimerFd() {
@@ -240,9 +271,20 @@ void Loop::drainPosted() {
 
 void Loop::run() {
   stopped_ = false;
+  // stall probe: with CPILOT_LOOP_DEBUG set, any single phase that
+  // holds the loop >10 ms is logged with its duration
+  const bool stallDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+  auto probe = [stallDebug](const char* phase, TimePoint t0) {
+    if (!stallDebug) return;
+    auto ms = std::chrono::duration<double, std::milli>(Clock::now() - t0)
+                  .count();
+    if (ms > 10.0) LOG_WARN("loop stall: %s took %.1f ms", phase, ms);
+  };
   std::vector<struct epoll_event> events(64);
   while (!stopped_) {
+    TimePoint t0 = Clock::now();
     drainDeferred();
+    probe("drainDeferred", t0);
     if (stopped_) break;
     armTimerFd();
     int timeoutMs = deferred_.empty() ? 1000 : 0;
@@ -258,10 +300,14 @@ void Loop::run() {
       if (it != fdCallbacks_.end()) {
         // copy: callback may unwatch itself
         FdCallback cb = it->second;
+        TimePoint tc = Clock::now();
         cb(events[i].events);
+        probe("fdCallback", tc);
       }
     }
+    TimePoint tt = Clock::now();
     fireDueTimers();
+    probe("fireDueTimers", tt);
   }
 }
 

@@ -8,6 +8,7 @@
 
 #include <cctype>
 #include <cstring>
+#include <cstdio>
 #include <map>
 
 #include "cpilot/log.hpp"
@@ -180,6 +181,16 @@ void Command::run(Loop& loop, std::shared_ptr<Bus> bus) {
 void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   loop_ = &loop;
   LOG_DEBUG("%s.Run start", name_.c_str());
+  static const bool spdbg = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+  auto tstart = Clock::now();
+  auto lap = [&](const char* what) {
+    if (!spdbg) return;
+    auto ms = std::chrono::duration<double, std::milli>(Clock::now() -
+                                                        tstart).count();
+    if (ms > 5.0)
+      fprintf(stderr, "spawn-step stall: %s after %.1f ms\n", what, ms);
+    tstart = Clock::now();
+  };
 
   int pipefds[2] = {-1, -1};
   if (!raw_) {
@@ -192,6 +203,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
     }
   }
 
+  lap("pipe2");
   running_ = true;
   pid_ = -1;
   pendingSignal_ = 0;
@@ -200,6 +212,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   // snapshot env (+ the PID-var overlay) on this (the loop) thread; the
   // spawner must never read the live environ concurrently with setenv
   auto baseEnv = baseEnvSnapshot();
+  lap("envSnapshot");
   // the spawner pool does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
   Spawner::global().spawn(
@@ -276,6 +289,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
           onExit(*loop_, bus, status);
         });
       });
+  lap("spawnerEnqueue");
 }
 
 void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {

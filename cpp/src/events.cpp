@@ -1,6 +1,7 @@
 #include "cpilot/events.hpp"
 
 #include <algorithm>
+#include <cstdio>
 
 #include "cpilot/log.hpp"
 
@@ -139,8 +140,8 @@ void Bus::drain() {
     dispatchHist_->observe(latency);
     static const bool latDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
     if (latDebug && latency > 0.004)
-      LOG_WARN("slow dispatch: %s waited %.1f ms (queue %zu)",
-               event.str().c_str(), latency * 1e3, queue_.size());
+      fprintf(stderr, "slow dispatch: %s waited %.1f ms (queue %zu)\n",
+              event.str().c_str(), latency * 1e3, queue_.size());
     // reservoir sample: bounded memory with uniform coverage of the
     // whole run (a plain prefix window stopped representing steady
     // state and grew ~16 KB/s until its cap)

@@ -6,6 +6,7 @@
 #include <sys/wait.h>
 #include <unistd.h>
 
+#include <cstdio>
 #include <cstring>
 #include <set>
 #include <stdexcept>
@@ -199,6 +200,18 @@ void Loop::armTimerFd() {
 }
 
 void Loop::fireDueTimers() {
+  static const bool stallDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+  auto timeCb = [](Timer* t, const TimerCallback& cb) {
+    if (!stallDebug) { cb(); return; }
+    TimePoint t0 = Clock::now();
+    cb();
+    auto ms = std::chrono::duration<double, std::milli>(Clock::now() - t0)
+                  .count();
+    if (ms > 10.0)
+      fprintf(stderr, "timer stall: id=%llu interval_ms=%lld took %.1f ms\n",
+              (unsigned long long)t->id,
+              (long long)(t->interval.count() / 1000000), ms);
+  };
   auto now = Clock::now();
   while (!timers_.empty()) {
     auto t = timers_.top();
@@ -220,10 +233,12 @@ void Loop::fireDueTimers() {
         t->deadline += Ns(periods * t->interval.count());
       }
       timers_.push(t);
-      t->cb();
+      timeCb(t.get(), t->cb);
     } else {
+      // keep t alive: erase drops the map reference
+      auto keep = t;
       timersById_.erase(t->id);
-      t->cb();
+      timeCb(keep.get(), keep->cb);
     }
   }
   armTimerFd();
@@ -274,11 +289,14 @@ void Loop::run() {
   // stall probe: with CPILOT_LOOP_DEBUG set, any single phase that
   // holds the loop >10 ms is logged with its duration
   const bool stallDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
-  auto probe = [stallDebug](const char* phase, TimePoint t0) {
+  auto probe = [stallDebug](const char* phase, TimePoint t0, int fd = -1) {
     if (!stallDebug) return;
     auto ms = std::chrono::duration<double, std::milli>(Clock::now() - t0)
                   .count();
-    if (ms > 10.0) LOG_WARN("loop stall: %s took %.1f ms", phase, ms);
+    // raw stderr: must bypass the configured log level (the stress
+    // shape runs at ERROR) and add no steady-state log traffic
+    if (ms > 10.0)
+      fprintf(stderr, "loop stall: %s fd=%d took %.1f ms\n", phase, fd, ms);
   };
   std::vector<struct epoll_event> events(64);
   while (!stopped_) {
@@ -302,7 +320,7 @@ void Loop::run() {
         FdCallback cb = it->second;
         TimePoint tc = Clock::now();
         cb(events[i].events);
-        probe("fdCallback", tc);
+        probe("fdCallback", tc, fd);
       }
     }
     TimePoint tt = Clock::now();

@@ -166,7 +166,13 @@ void Spawner::spawn(Loop& loop, std::string execPath,
     for (auto& e : *baseEnv) append(e);
   for (auto& e : extraEnv) append(e);
 
+  TimePoint tLock = Clock::now();
   std::lock_guard<std::mutex> l(mu_);
+  if (spawnDebug()) {
+    auto ms = std::chrono::duration<double, std::milli>(Clock::now() -
+                                                        tLock).count();
+    if (ms > 5.0) fprintf(stderr, "spawner mu_ wait %.1f ms\n", ms);
+  }
   uint32_t reqId = nextReqId_++;
   hdr.reqId = reqId;
   memcpy(buf.data(), &hdr, sizeof(hdr));

@@ -18,10 +18,12 @@ port = free_port()
 cfg = stress_config("localhost:79", port, jobs, 0, 100,
                     os.path.join(wd, "cp.socket"))
 d = harness.Daemon(config_dict=cfg, workdir=wd,
-                   env={"CPILOT_LOOP_DEBUG": "1"})
+                   env={"CPILOT_LOOP_DEBUG": "1", "CPILOT_SPAWN_DEBUG": "0"})
+
 d.start()
 d.wait_for_socket(timeout=60)
 time.sleep(6)
+marker = len(d.log())
 s0 = scrape(port)
 t0 = time.time()
 time.sleep(15)
@@ -29,8 +31,11 @@ s1 = scrape(port)
 el = time.time() - t0
 pub = (s1["published"] - s0["published"]) / el
 p99 = histogram_p99(s0["buckets"], s1["buckets"])
-log = d.log()
-stalls = re.findall(r"loop stall: (\w+) took ([0-9.]+) ms", log)
+log = d.log()[marker:]
+stalls = re.findall(r"loop stall: (\w+) fd=-?\d+ took ([0-9.]+) ms", log)
+tstalls = re.findall(r"timer stall: id=\d+ interval_ms=(\d+) took ([0-9.]+) ms", log)
+sstalls = re.findall(r"spawn-step stall: (\w+) after ([0-9.]+) ms", log)
+muw = re.findall(r"spawner mu_ wait ([0-9.]+) ms", log)
 agg = collections.Counter()
 mx = collections.defaultdict(float)
 tot = collections.defaultdict(float)
@@ -49,5 +54,10 @@ print(json.dumps({
     "stall_total_ms": {k: round(v, 1) for k, v in tot.items()},
     "slow_dispatch_count": len(slow), "slow_codes": dict(codes),
     "slow_worst_ms": worst,
+    "timer_stalls": len(tstalls),
+    "timer_worst_ms": sorted((float(m) for _, m in tstalls))[-5:],
+    "spawn_step_stalls": collections.Counter(w for w, _ in sstalls),
+    "spawn_step_worst_ms": sorted((float(m) for _, m in sstalls))[-5:],
+    "mu_waits": len(muw),
 }), flush=True)
 d.cleanup()

@@ -3,6 +3,8 @@
 // Parity: /root/reference/config/logger/logging.go:39-129.
 #pragma once
 
+#include <cstdlib>
+
 #include <cstdarg>
 #include <string>
 
@@ -40,4 +42,12 @@ void logFields(Level l, const std::string& job, int pid, const std::string& msg)
 #define LOG_ERROR(...) ::cpilot::logging::logf(::cpilot::logging::Level::Error, __VA_ARGS__)
 
 }  // namespace logging
+}  // namespace cpilot
+
+namespace cpilot {
+// env flag helper: set and not "0"/"" means on
+inline bool cpilotDebugEnv(const char* name) {
+  const char* v = getenv(name);
+  return v && v[0] && !(v[0] == '0' && !v[1]);
+}
 }  // namespace cpilot

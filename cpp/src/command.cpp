@@ -181,7 +181,7 @@ void Command::run(Loop& loop, std::shared_ptr<Bus> bus) {
 void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   loop_ = &loop;
   LOG_DEBUG("%s.Run start", name_.c_str());
-  static const bool spdbg = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+  static const bool spdbg = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
   auto tstart = Clock::now();
   auto lap = [&](const char* what) {
     if (!spdbg) return;

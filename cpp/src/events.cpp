@@ -138,7 +138,7 @@ void Bus::drain() {
     double latency =
         std::chrono::duration<double>(Clock::now() - publishedAt).count();
     dispatchHist_->observe(latency);
-    static const bool latDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+    static const bool latDebug = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
     if (latDebug && latency > 0.004)
       fprintf(stderr, "slow dispatch: %s waited %.1f ms (queue %zu)\n",
               event.str().c_str(), latency * 1e3, queue_.size());

@@ -200,7 +200,7 @@ void Loop::armTimerFd() {
 }
 
 void Loop::fireDueTimers() {
-  static const bool stallDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+  static const bool stallDebug = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
   auto timeCb = [](Timer* t, const TimerCallback& cb) {
     if (!stallDebug) { cb(); return; }
     TimePoint t0 = Clock::now();
@@ -288,7 +288,7 @@ void Loop::run() {
   stopped_ = false;
   // stall probe: with CPILOT_LOOP_DEBUG set, any single phase that
   // holds the loop >10 ms is logged with its duration
-  const bool stallDebug = getenv("CPILOT_LOOP_DEBUG") != nullptr;
+  const bool stallDebug = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
   auto probe = [stallDebug](const char* phase, TimePoint t0, int fd = -1) {
     if (!stallDebug) return;
     auto ms = std::chrono::duration<double, std::milli>(Clock::now() - t0)

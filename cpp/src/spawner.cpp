@@ -53,7 +53,7 @@ std::string findHelperBinary() {
 }
 
 bool spawnDebug() {
-  static bool on = getenv("CPILOT_SPAWN_DEBUG") != nullptr;
+  static bool on = cpilotDebugEnv("CPILOT_SPAWN_DEBUG");
   return on;
 }
 

@@ -32,7 +32,8 @@ namespace {
 int selfPipe[2] = {-1, -1};
 
 bool helperDebug() {
-  static bool on = getenv("CPILOT_SPAWN_DEBUG") != nullptr;
+  const char* v = getenv("CPILOT_SPAWN_DEBUG");
+  static bool on = v && v[0] && !(v[0] == '0' && !v[1]);
   return on;
 }
 

@@ -139,8 +139,14 @@ std::string mapToJson(const std::map<std::string, std::string>& m) {
 }  // namespace
 
 int main(int argc, char** argv) {
-  // PID-1: fork into the supervisor before anything else (main.go:25-28)
-  if (getpid() == 1) {
+  // PID-1: fork into the supervisor before anything else (main.go:25-28).
+  // CPILOT_FORCE_SUP forces the same split without PID 1 (sup becomes a
+  // child subreaper): for hosts whose init doesn't reap, and for
+  // unprivileged tests of the reap path (no PID namespace needed).
+  const char* forceSup = getenv("CPILOT_FORCE_SUP");
+  if (getpid() == 1 ||
+      (forceSup && forceSup[0] && !(forceSup[0] == '0' && !forceSup[1]))) {
+    unsetenv("CPILOT_FORCE_SUP");  // the worker must not recurse
     int rc = supRun(argc, argv);
     if (rc >= 0) return rc;  // parent (supervisor) path
     // child falls through as the worker

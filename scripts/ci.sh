@@ -4,6 +4,7 @@
 # pipeline (.travis.yml:22-26) without docker.
 set -e
 cd "$(dirname "$0")/.."
+./scripts/lint.sh
 make build
 ./bin/cpilot_unittests
 python3 -m pytest tests/ -q -m "not gpu"

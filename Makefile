@@ -44,7 +44,7 @@ soak: build
 capacity: build
 	python3 scripts/capacity.py
 
-VERSION := 3.9.0-amd
+VERSION := 3.10.0-amd
 release: build
 	mkdir -p release
 	tar -czf release/containerpilot-$(VERSION).tar.gz -C bin containerpilot cpilot-spawn-helper

@@ -5,7 +5,7 @@
 namespace cpilot {
 
 #ifndef CPILOT_VERSION
-#define CPILOT_VERSION "3.9.0-amd"
+#define CPILOT_VERSION "3.10.0-amd"
 #endif
 #ifndef CPILOT_GITHASH
 #define CPILOT_GITHASH ""

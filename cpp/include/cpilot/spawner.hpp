@@ -52,6 +52,14 @@ class Spawner {
 
   int helperCount() const { return (int)helpers_.size(); }
 
+  // Requests in flight (sent or queued, not yet replied). Used for
+  // EARLY shedding: a health-check tick that would join a deep backlog
+  // is skipped at tick time (same observable as the single-instance
+  // skip — a shed check — but it keeps queues and round trips short,
+  // so saturation degrades flat instead of collapsing).
+  size_t backlog();
+  bool overloaded();
+
  private:
   Spawner();
   struct Helper {

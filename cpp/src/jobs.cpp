@@ -7,6 +7,7 @@
 #include "cpilot/decode.hpp"
 #include "cpilot/ips.hpp"
 #include "cpilot/log.hpp"
+#include "cpilot/spawner.hpp"
 
 namespace cpilot {
 
@@ -701,7 +702,14 @@ Job::HandleResult Job::onHeartbeatTimerExpired() {
   JobStatus status = getStatus();
   if (status != JobStatus::Maintenance && status != JobStatus::Idle) {
     if (healthCheckExec_) {
-      healthCheckExec_->run(*loop_, bus_);
+      if (Spawner::global().overloaded()) {
+        // early shed: joining a deep spawn backlog only inflates
+        // round trips; skipping now is the same observable as the
+        // single-instance skip (an overloaded daemon sheds checks)
+        LOG_DEBUG("%s check shed: spawn backlog deep", name_.c_str());
+      } else {
+        healthCheckExec_->run(*loop_, bus_);
+      }
     } else if (service_) {
       // non-checked but advertised services (telemetry endpoint)
       sendHeartbeat();

@@ -123,6 +123,17 @@ bool Spawner::launchHelper(Helper* h) {
   return true;
 }
 
+size_t Spawner::backlog() {
+  std::lock_guard<std::mutex> l(mu_);
+  return pending_.size();
+}
+
+bool Spawner::overloaded() {
+  // ~16 outstanding per helper ≈ a few ms of queue at measured spawn
+  // cost; beyond that a new check would only inflate round trips
+  return backlog() > helpers_.size() * 16;
+}
+
 void Spawner::wakeReader() {
   char b = 1;
   ssize_t unused = write(wakeFds_[1], &b, 1);

@@ -26,6 +26,11 @@ using Clock = std::chrono::steady_clock;
 using TimePoint = Clock::time_point;
 using Ns = std::chrono::nanoseconds;
 
+// Reset the calling thread to default scheduling (SCHED_OTHER, nice 0).
+// The daemon's reactor thread runs elevated (SCHED_RR or negative nice);
+// background threads and child processes must not inherit that.
+void resetThreadScheduling();
+
 class Loop {
  public:
   using FdCallback = std::function<void(uint32_t epollEvents)>;

@@ -195,6 +195,7 @@ void ConsulBackend::stop() {
 }
 
 void ConsulBackend::workerMain() {
+  resetThreadScheduling();  // do not inherit the reactor's RT priority
   while (true) {
     std::function<void()> task;
     {
@@ -421,6 +422,7 @@ void ConsulBackend::healthServiceBlocking(const std::string& name,
   // may outlive the backend and touches only the shared reg + the loop
   std::thread([reg, target, tokenHdr, tls, path, cb, loop, token,
                waitSeconds] {
+    resetThreadScheduling();
     std::map<std::string, std::string> headers;
     if (!tokenHdr.empty()) headers["X-Consul-Token"] = tokenHdr;
     auto res = http::request(target, "GET", path, "", "application/json",

@@ -1,6 +1,8 @@
 #include "cpilot/loop.hpp"
 
 #include <fcntl.h>
+#include <sched.h>
+#include <sys/resource.h>
 #include <sys/epoll.h>
 #include <sys/timerfd.h>
 #include <sys/wait.h>
@@ -14,6 +16,13 @@
 #include "cpilot/log.hpp"
 
 namespace cpilot {
+
+void resetThreadScheduling() {
+  struct sched_param sp;
+  memset(&sp, 0, sizeof(sp));
+  sched_setscheduler(0, SCHED_OTHER, &sp);
+  setpriority(PRIO_PROCESS, 0, 0);
+}
 
 namespace {
 // registry of live loops for postIfLive (guards cross-thread posts that

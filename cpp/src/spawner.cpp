@@ -248,6 +248,7 @@ bool Spawner::sendRequest(Helper& h, const std::vector<char>& buf, int fd) {
 }
 
 void Spawner::readerMain() {
+  resetThreadScheduling();  // do not inherit the reactor's RT priority
   std::vector<char> buf(64);
   while (true) {
     std::vector<struct pollfd> fds;

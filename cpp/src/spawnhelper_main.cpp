@@ -13,7 +13,9 @@
 #include <fcntl.h>
 #include <poll.h>
 #include <signal.h>
+#include <sched.h>
 #include <spawn.h>
+#include <sys/resource.h>
 #include <stdio.h>
 #include <stdlib.h>
 #include <string.h>
@@ -126,6 +128,14 @@ void handleRequest(int sock, const char* buf, size_t len, int stdioFd) {
 }  // namespace
 
 int cpilotSpawnHelperMain() {
+  // the daemon's reactor may run SCHED_RR / negative nice; neither the
+  // helper nor the processes it spawns should inherit that
+  {
+    struct sched_param sp;
+    memset(&sp, 0, sizeof(sp));
+    sched_setscheduler(0, SCHED_OTHER, &sp);
+    setpriority(PRIO_PROCESS, 0, 0);
+  }
   const int sock = kHelperFd;
   // nonblocking: the drain loop must hit EAGAIN, not block, once the
   // pending wakeup bytes are consumed

@@ -262,6 +262,12 @@ void Spawner::readerMain() {
     }
     fds.push_back({wakeFds_[0], POLLIN, 0});
     int rc = poll(fds.data(), (nfds_t)fds.size(), 1000);
+    if (spawnDebug() && rc != 0) {
+      fprintf(stderr, "[spawner] poll rc=%d", rc);
+      for (size_t i = 0; i < fds.size(); i++)
+        fprintf(stderr, " fd%d=%x", fds[i].fd, fds[i].revents);
+      fprintf(stderr, "\n");
+    }
     if (rc < 0) {
       if (errno == EINTR) continue;
       return;

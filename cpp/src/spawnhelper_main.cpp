@@ -137,6 +137,12 @@ int cpilotSpawnHelperMain() {
     setpriority(PRIO_PROCESS, 0, 0);
   }
   const int sock = kHelperFd;
+  // The dup2 that placed the socket at fd 3 cleared CLOEXEC so it
+  // survived OUR exec; restore it now or every process this helper
+  // spawns inherits the control socket — which both leaks an fd into
+  // user processes and keeps the socketpair alive after this helper
+  // dies, so the daemon would never see EOF and never respawn it.
+  fcntl(sock, F_SETFD, FD_CLOEXEC);
   // nonblocking: the drain loop must hit EAGAIN, not block, once the
   // pending wakeup bytes are consumed
   if (pipe2(selfPipe, O_NONBLOCK | O_CLOEXEC) != 0) _exit(1);

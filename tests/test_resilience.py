@@ -148,3 +148,64 @@ def test_bind_retry_does_not_block_dispatch(daemon_factory):
     assert b"ticker" in body
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_spawn_helper_death_recovers(daemon_factory):
+    """Killing a spawn helper mid-run is survived: the daemon respawns
+    it, in-flight launches fail loudly (ExitFailed -> restart policy),
+    and health checks keep completing afterwards."""
+    import os
+    import signal as signalmod
+    import time
+
+    d = daemon_factory({
+        "consul": "127.0.0.1:1",
+        "stopTimeout": 1,
+        "logging": {"level": "DEBUG"},
+        "jobs": [
+            {"name": "main-app", "exec": ["sleep", "60"],
+             "port": 23456, "interfaces": ["static:127.0.0.1"],
+             "health": {"exec": ["true"], "interval": "500ms", "ttl": 5}},
+        ],
+    }).start()
+    d.wait_for_socket()
+    time.sleep(1.5)
+
+    # helpers are direct children of the worker process named
+    # cpilot-spawn-helper; the worker is d.proc.pid (no sup here)
+    def helper_pids():
+        pids = []
+        for pid in os.listdir("/proc"):
+            if not pid.isdigit():
+                continue
+            try:
+                with open("/proc/%s/stat" % pid) as f:
+                    parts = f.read().split()
+                if parts[1] == "(cpilot-spawn-he)" and \
+                        int(parts[3]) == d.proc.pid:
+                    pids.append(int(pid))
+            except (OSError, ValueError):
+                pass
+        return pids
+
+    before = helper_pids()
+    assert before, "no spawn helpers found under the daemon"
+    os.kill(before[0], signalmod.SIGKILL)
+
+    # respawn logged, and checks still complete after the death
+    deadline = time.time() + 10
+    while time.time() < deadline and "died; respawning" not in d.log():
+        time.sleep(0.2)
+    assert "died; respawning" in d.log(), d.log()[-2000:]
+    marker = d.log().count("check.main-app exited without error")
+    deadline = time.time() + 10
+    while time.time() < deadline and \
+            d.log().count("check.main-app exited without error") < marker + 3:
+        time.sleep(0.2)
+    assert d.log().count("check.main-app exited without error") >= marker + 3, \
+        "checks did not resume after helper death:\n" + d.log()[-2000:]
+    # pool healed: same helper count, new pid present
+    after = helper_pids()
+    assert len(after) == len(before), (before, after)
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -483,6 +483,55 @@ struct Transport {
 
 }  // namespace
 
+namespace {
+
+// Thread-local keep-alive pool: one cached plain-TCP/unix connection
+// per target, giving the same connection reuse the reference's consul
+// client inherits from Go's http.Transport. Plain + uncancellable
+// requests only (TLS sessions and cancellable long-polls stay
+// per-request). At thousands of TTL updates/sec, per-request connects
+// burned measurable CPU on both the daemon and the agent.
+thread_local std::map<std::string, int>* connPool = nullptr;
+
+int poolTake(const std::string& target) {
+  if (!connPool) return -1;
+  auto it = connPool->find(target);
+  if (it == connPool->end()) return -1;
+  int fd = it->second;
+  connPool->erase(it);
+  return fd;
+}
+
+void poolStore(const std::string& target, int fd) {
+  if (!connPool) connPool = new std::map<std::string, int>();
+  auto it = connPool->find(target);
+  if (it != connPool->end()) {
+    close(it->second);
+    it->second = fd;
+  } else {
+    (*connPool)[target] = fd;
+  }
+}
+
+// incremental chunked-body decoder: returns true once the terminal
+// chunk is complete, filling *out with the decoded bytes
+bool tryDechunk(const std::string& data, std::string* out) {
+  out->clear();
+  size_t pos = 0;
+  while (true) {
+    size_t eol = data.find("\r\n", pos);
+    if (eol == std::string::npos) return false;
+    long len = strtol(data.substr(pos, eol - pos).c_str(), nullptr, 16);
+    if (len < 0) return false;
+    if (len == 0) return true;  // terminal chunk (ignore trailers)
+    if (data.size() < eol + 2 + (size_t)len + 2) return false;
+    out->append(data, eol + 2, (size_t)len);
+    pos = eol + 2 + (size_t)len + 2;
+  }
+}
+
+}  // namespace
+
 ClientResult request(const std::string& target, const std::string& method,
                      const std::string& path, const std::string& body,
                      const std::string& contentType,
@@ -494,99 +543,170 @@ ClientResult request(const std::string& target, const std::string& method,
     result.error = "cancelled";
     return result;
   }
-  Transport t;
-  t.fd = connectTarget(target, timeoutMs, &result.error);
-  if (t.fd < 0) return result;
-  struct Disarm {
-    CancelToken* ct;
-    ~Disarm() {
-      if (ct) ct->disarm();
+  const bool tlsOn = tls && tls->enabled;
+  const bool reusable = !tlsOn && cancel == nullptr;
+
+  for (int attempt = 0; attempt < 2; attempt++) {
+    result = ClientResult{};
+    Transport t;
+    bool fromPool = false;
+    if (reusable && attempt == 0) {
+      t.fd = poolTake(target);
+      fromPool = t.fd >= 0;
     }
-  } disarm{cancel};
-  if (cancel) cancel->arm(t.fd);
+    if (t.fd < 0) {
+      t.fd = connectTarget(target, timeoutMs, &result.error);
+      if (t.fd < 0) return result;
+    } else {
+      // refresh the timeouts: the pooled fd carries the previous
+      // request's deadline settings
+      struct timeval tv;
+      tv.tv_sec = timeoutMs / 1000;
+      tv.tv_usec = (timeoutMs % 1000) * 1000;
+      setsockopt(t.fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+      setsockopt(t.fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
+    }
+    struct Disarm {
+      CancelToken* ct;
+      ~Disarm() {
+        if (ct) ct->disarm();
+      }
+    } disarm{cancel};
+    if (cancel) cancel->arm(t.fd);
 
-  std::string host = target.rfind("unix:", 0) == 0 ? "localhost" : target;
-  if (tls && tls->enabled) {
-    if (!t.startTls(host, *tls, &result.error)) return result;
-  }
+    std::string host = target.rfind("unix:", 0) == 0 ? "localhost" : target;
+    if (tlsOn) {
+      if (!t.startTls(host, *tls, &result.error)) return result;
+    }
 
-  std::string req = method + " " + path + " HTTP/1.1\r\n";
-  req += "Host: " + host + "\r\n";
-  req += "Connection: close\r\n";
-  for (auto& kv : headers) req += kv.first + ": " + kv.second + "\r\n";
-  if (!body.empty() || method == "POST" || method == "PUT") {
-    req += "Content-Type: " + contentType + "\r\n";
-    req += "Content-Length: " + std::to_string(body.size()) + "\r\n";
-  }
-  req += "\r\n";
-  req += body;
+    std::string req = method + " " + path + " HTTP/1.1\r\n";
+    req += "Host: " + host + "\r\n";
+    req += reusable ? "Connection: keep-alive\r\n" : "Connection: close\r\n";
+    for (auto& kv : headers) req += kv.first + ": " + kv.second + "\r\n";
+    if (!body.empty() || method == "POST" || method == "PUT") {
+      req += "Content-Type: " + contentType + "\r\n";
+      req += "Content-Length: " + std::to_string(body.size()) + "\r\n";
+    }
+    req += "\r\n";
+    req += body;
 
-  size_t off = 0;
-  while (off < req.size()) {
-    ssize_t n = t.send(req.data() + off, req.size() - off);
-    if (n <= 0) {
-      if (n < 0 && errno == EINTR && !t.ssl) continue;
+    bool ioFailed = false;
+    size_t off = 0;
+    while (off < req.size()) {
+      ssize_t n = t.send(req.data() + off, req.size() - off);
+      if (n <= 0) {
+        if (n < 0 && errno == EINTR && !t.ssl) continue;
+        ioFailed = true;
+        break;
+      }
+      off += n;
+    }
+    if (ioFailed) {
+      if (fromPool) continue;  // stale keep-alive: retry on a fresh conn
       result.error = "write failed";
       return result;
     }
-    off += n;
-  }
 
-  std::string resp;
-  char buf[8192];
-  while (true) {
-    ssize_t n = t.recv(buf, sizeof(buf));
-    if (n > 0) {
-      resp.append(buf, n);
-    } else if (n < 0 && errno == EINTR && !t.ssl) {
-      continue;
-    } else {
-      break;
+    // read incrementally: headers first, then exactly the framed body
+    std::string resp;
+    char buf[8192];
+    size_t headerEnd = std::string::npos;
+    auto readMore = [&]() -> bool {
+      while (true) {
+        ssize_t n = t.recv(buf, sizeof(buf));
+        if (n > 0) {
+          resp.append(buf, n);
+          return true;
+        }
+        if (n < 0 && errno == EINTR && !t.ssl) continue;
+        return false;  // EOF or error/timeout
+      }
+    };
+    bool eof = false;
+    while ((headerEnd = resp.find("\r\n\r\n")) == std::string::npos) {
+      if (!readMore()) {
+        eof = true;
+        break;
+      }
     }
-  }
+    if (headerEnd == std::string::npos || resp.compare(0, 5, "HTTP/") != 0) {
+      if (fromPool && resp.empty()) continue;  // server closed the idle conn
+      result.error = (cancel && cancel->cancelled()) ? "cancelled"
+                                                     : "malformed response";
+      return result;
+    }
 
-  size_t headerEnd = resp.find("\r\n\r\n");
-  if (headerEnd == std::string::npos || resp.compare(0, 5, "HTTP/") != 0) {
-    result.error = (cancel && cancel->cancelled()) ? "cancelled"
-                                                   : "malformed response";
+    size_t sp = resp.find(' ');
+    result.status = atoi(resp.c_str() + sp + 1);
+    // response headers (lower-cased keys)
+    size_t pos = resp.find("\r\n") + 2;
+    while (pos < headerEnd) {
+      size_t eol = resp.find("\r\n", pos);
+      if (eol == std::string::npos || eol > headerEnd) break;
+      std::string line = resp.substr(pos, eol - pos);
+      pos = eol + 2;
+      size_t colon = line.find(':');
+      if (colon == std::string::npos) continue;
+      std::string key = line.substr(0, colon);
+      for (auto& ch : key) ch = (char)tolower((unsigned char)ch);
+      size_t vstart = colon + 1;
+      while (vstart < line.size() && line[vstart] == ' ') vstart++;
+      result.headers[key] = line.substr(vstart);
+    }
+
+    auto lower = [](std::string v) {
+      for (auto& ch : v) ch = (char)tolower((unsigned char)ch);
+      return v;
+    };
+    bool chunked = false;
+    auto te = result.headers.find("transfer-encoding");
+    if (te != result.headers.end() &&
+        lower(te->second).find("chunked") != std::string::npos)
+      chunked = true;
+    bool bodyFramed = true;
+    if (chunked) {
+      std::string decoded;
+      while (!tryDechunk(resp.substr(headerEnd + 4), &decoded)) {
+        if (eof || !readMore()) {
+          bodyFramed = false;
+          break;
+        }
+      }
+      result.body = decoded;
+    } else if (result.headers.count("content-length")) {
+      size_t want =
+          (size_t)atoll(result.headers["content-length"].c_str());
+      while (resp.size() < headerEnd + 4 + want) {
+        if (eof || !readMore()) {
+          bodyFramed = false;
+          break;
+        }
+      }
+      result.body = resp.substr(headerEnd + 4,
+                                std::min(want, resp.size() - headerEnd - 4));
+    } else {
+      // no framing: read to EOF, connection not reusable
+      while (readMore()) {
+      }
+      result.body = resp.substr(headerEnd + 4);
+      bodyFramed = false;
+    }
+
+    result.ok = true;
+    // reuse only when both sides agreed and the body was fully framed
+    bool serverKeeps = true;
+    auto ch = result.headers.find("connection");
+    if (ch != result.headers.end() &&
+        lower(ch->second).find("close") != std::string::npos)
+      serverKeeps = false;
+    if (reusable && bodyFramed && serverKeeps && result.status > 0) {
+      int fd = t.fd;
+      t.fd = -1;  // keep it out of Transport's destructor
+      poolStore(target, fd);
+    }
     return result;
   }
-  size_t sp = resp.find(' ');
-  result.status = atoi(resp.c_str() + sp + 1);
-  result.body = resp.substr(headerEnd + 4);
-  // response headers (lower-cased keys)
-  size_t pos = resp.find("\r\n") + 2;
-  while (pos < headerEnd) {
-    size_t eol = resp.find("\r\n", pos);
-    if (eol == std::string::npos || eol > headerEnd) break;
-    std::string line = resp.substr(pos, eol - pos);
-    pos = eol + 2;
-    size_t colon = line.find(':');
-    if (colon == std::string::npos) continue;
-    std::string key = line.substr(0, colon);
-    for (auto& ch : key) ch = tolower((unsigned char)ch);
-    size_t vstart = colon + 1;
-    while (vstart < line.size() && line[vstart] == ' ') vstart++;
-    result.headers[key] = line.substr(vstart);
-  }
-  // chunked responses: dechunk (Consul uses Content-Length, but be safe)
-  std::string lower = resp.substr(0, headerEnd);
-  for (auto& ch : lower) ch = tolower((unsigned char)ch);
-  if (lower.find("transfer-encoding: chunked") != std::string::npos) {
-    std::string out;
-    size_t cpos = 0;
-    const std::string& cb = result.body;
-    while (cpos < cb.size()) {
-      size_t eol = cb.find("\r\n", cpos);
-      if (eol == std::string::npos) break;
-      long len = strtol(cb.substr(cpos, eol - cpos).c_str(), nullptr, 16);
-      if (len <= 0) break;
-      out += cb.substr(eol + 2, len);
-      cpos = eol + 2 + len + 2;
-    }
-    result.body = out;
-  }
-  result.ok = true;
+  result.error = "write failed";
   return result;
 }
 

@@ -130,6 +130,27 @@ def histogram_p99(b0, b1):
     return None
 
 
+def cgroup_cpu_info():
+    """(quota_cores, throttled_usec) from cgroup v2, or (None, None)."""
+    quota = None
+    throttled = None
+    try:
+        with open("/sys/fs/cgroup/cpu.max") as f:
+            parts = f.read().split()
+        if parts[0] != "max":
+            quota = round(float(parts[0]) / float(parts[1]), 1)
+    except (OSError, ValueError, IndexError):
+        pass
+    try:
+        with open("/sys/fs/cgroup/cpu.stat") as f:
+            for line in f:
+                if line.startswith("throttled_usec"):
+                    throttled = int(line.split()[1])
+    except (OSError, ValueError):
+        pass
+    return quota, throttled
+
+
 def wait_http(port, timeout=30):
     deadline = time.time() + timeout
     while time.time() < deadline:
@@ -198,12 +219,16 @@ def main():
                 torch.cuda.synchronize()
 
         sync()
+        quota_cores, throttled0 = cgroup_cpu_info()
         t0 = time.monotonic()
         s0 = scrape(port)
         time.sleep(args.steps)
         sync()
         s1 = scrape(port)
         t1 = time.monotonic()
+        _, throttled1 = cgroup_cpu_info()
+        throttled_ms = (None if throttled0 is None or throttled1 is None
+                        else round((throttled1 - throttled0) / 1000.0, 1))
 
         elapsed = t1 - t0
         published = (s1["published"] - s0["published"]) / elapsed
@@ -260,6 +285,11 @@ def main():
                     "delivered_events_per_sec": round(total_delivered, 1),
                     "check_completion_pct": round(completion_pct, 1),
                     "p99_dispatch_ms": round(worst_p99_ms, 4),
+                    # cgroup CPU budget + throttling during the window:
+                    # a quota-throttled run (shared slice too small for
+                    # N ranks of real forked checks) self-documents here
+                    "cgroup_cpu_quota_cores": quota_cores,
+                    "cgroup_throttled_ms_in_window": throttled_ms,
                     "p99_target_ms": 1.0,
                     "events_per_sec_target": BASELINE_EVENTS_PER_SEC,
                 },

@@ -67,6 +67,7 @@ class Command : public std::enable_shared_from_this<Command> {
   int pendingSignal_ = 0;  // term/kill requested while spawn in flight
   Loop* loop_ = nullptr;
   uint64_t timeoutTimer_ = 0;
+  TimePoint tReq_{}, tCb_{};  // spawn-RTT probe (CPILOT_LOOP_DEBUG)
   int logFd_ = -1;
   std::string logBuf_;
 };

@@ -207,6 +207,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   running_ = true;
   pid_ = -1;
   pendingSignal_ = 0;
+  tReq_ = Clock::now();
   auto self = shared_from_this();
   int readFd = pipefds[0];
   // snapshot env (+ the PID-var overlay) on this (the loop) thread; the
@@ -230,6 +231,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
           return;
         }
         pid_ = pid;
+        tCb_ = Clock::now();
         if (pidEnvName_.empty())
           pidEnvName_ = "CONTAINERPILOT_" + envName() + "_PID";
         overlaySet(pidEnvName_, std::to_string(pid));
@@ -294,6 +296,20 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
 
 void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
   LOG_DEBUG("%s.Run end", name_.c_str());
+  static const bool rttDebug = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
+  if (rttDebug) {
+    auto now = Clock::now();
+    double total =
+        std::chrono::duration<double, std::milli>(now - tReq_).count();
+    if (total > 50.0) {
+      double execPhase =
+          std::chrono::duration<double, std::milli>(now - tCb_).count();
+      fprintf(stderr,
+              "spawn rtt: %s total=%.0fms spawn_phase=%.0fms "
+              "exec_phase=%.0fms\n",
+              name_.c_str(), total, total - execPhase, execPhase);
+    }
+  }
   if (timeoutTimer_) {
     loop.cancelTimer(timeoutTimer_);
     timeoutTimer_ = 0;

@@ -45,6 +45,9 @@ for ph, ms in stalls:
     tot[ph] += float(ms)
 slow = re.findall(r"slow dispatch: \{(\w+) [^}]*\} waited ([0-9.]+) ms"
                   r" \(queue (\d+)\)", log)
+rtts = [(float(a), float(b), float(c)) for a, b, c in re.findall(
+    r"spawn rtt: \S+ total=([0-9.]+)ms spawn_phase=([0-9.]+)ms"
+    r" exec_phase=([0-9.]+)ms", log)]
 codes = collections.Counter(c for c, _, _ in slow)
 worst = sorted((float(ms) for _, ms, _ in slow), reverse=True)[:10]
 print(json.dumps({
@@ -59,5 +62,11 @@ print(json.dumps({
     "spawn_step_stalls": collections.Counter(w for w, _ in sstalls),
     "spawn_step_worst_ms": sorted((float(m) for _, m in sstalls))[-5:],
     "mu_waits": len(muw),
+    "slow_rtts": len(rtts),
+    "rtt_samples": sorted(rtts, reverse=True)[:8],
+    "rtt_mean_spawn_phase": round(sum(r[1] for r in rtts) /
+                                  max(1, len(rtts)), 1),
+    "rtt_mean_exec_phase": round(sum(r[2] for r in rtts) /
+                                 max(1, len(rtts)), 1),
 }), flush=True)
 d.cleanup()

@@ -125,14 +125,16 @@ void Bus::publish(Event event) {
 
 void Bus::drain() {
   drainScheduled_ = false;
-  // process what's queued now (bounded: a completion burst can enqueue
-  // hundreds of events; interleave timer firing between batches so
-  // dispatch latency stays bounded — publishes from handlers and the
-  // remainder go to the next batch, order preserved)
-  constexpr size_t kDrainBatch = 64;
+  // process what's queued now, time-budgeted: a completion burst can
+  // enqueue hundreds of events and unbounded drains stall timers, but
+  // a fixed count caps sustained throughput at count x iterations/s
+  // (measured throttling the 2000-job shape). ~2 ms of dispatch per
+  // drain, remainder re-deferred, order preserved.
+  constexpr auto kBudget = std::chrono::milliseconds(2);
+  TimePoint start = Clock::now();
   size_t n = queue_.size();
-  if (n > kDrainBatch) n = kDrainBatch;
   for (size_t i = 0; i < n && !queue_.empty(); i++) {
+    if ((i & 63) == 63 && Clock::now() - start >= kBudget) break;
     auto [event, publishedAt] = std::move(queue_.front());
     queue_.pop_front();
     double latency =

@@ -265,27 +265,33 @@ void Loop::drainDeferred() {
 }
 
 void Loop::drainPosted() {
-  // Bounded batch for the same reason as reapChildren: spawn
-  // completions posted by the spawner pool each do real setup work
-  // (pipe watch, timeout timer, child watch — possibly a synchronous
-  // exit). Leftovers re-arm the wakeup pipe so they run next iteration
-  // with timers interleaved.
-  constexpr size_t kPostedBatch = 64;
-  std::deque<std::function<void()>> batch;
-  bool more = false;
-  {
-    std::lock_guard<std::mutex> l(postedMu_);
-    if (posted_.size() <= kPostedBatch) {
-      batch.swap(posted_);
-    } else {
-      for (size_t i = 0; i < kPostedBatch; i++) {
+  // Time-budgeted batch: spawn completions/exit notices each do real
+  // setup work, so draining an unbounded burst stalls timers — but a
+  // FIXED count caps sustained throughput at count x iterations/s,
+  // which was measured to throttle the 2000-job shape (per-iteration
+  // tick work grows with jobs, iteration rate falls, and the posted
+  // queue backed up ~330 ms). Process up to ~2 ms of work, then
+  // re-arm the wakeup pipe for the remainder so timers interleave.
+  constexpr auto kBudget = std::chrono::milliseconds(2);
+  constexpr size_t kHardCap = 4096;
+  TimePoint start = Clock::now();
+  size_t processed = 0;
+  bool more = true;
+  while (more && processed < kHardCap &&
+         Clock::now() - start < kBudget) {
+    std::deque<std::function<void()>> batch;
+    {
+      std::lock_guard<std::mutex> l(postedMu_);
+      for (size_t i = 0; i < 64 && !posted_.empty(); i++) {
         batch.push_back(std::move(posted_.front()));
         posted_.pop_front();
       }
-      more = true;
+      more = !posted_.empty();
     }
+    for (auto& fn : batch) fn();
+    processed += batch.size();
+    if (batch.empty()) break;
   }
-  for (auto& fn : batch) fn();
   if (more) {
     char b = 1;
     ssize_t unused = write(wakeupFds_[1], &b, 1);

@@ -314,19 +314,31 @@ void Loop::run() {
       fprintf(stderr, "loop stall: %s fd=%d took %.1f ms\n", phase, fd, ms);
   };
   std::vector<struct epoll_event> events(64);
+  // phase accounting (stallDebug): 5 s summaries of where loop time goes
+  double accDeferred = 0, accEpoll = 0, accFds = 0, accTimers = 0;
+  uint64_t iters = 0;
+  TimePoint lastReport = Clock::now();
+  auto acc = [](double* slot, TimePoint t0) {
+    *slot += std::chrono::duration<double, std::milli>(Clock::now() - t0)
+                 .count();
+  };
   while (!stopped_) {
     TimePoint t0 = Clock::now();
     drainDeferred();
     probe("drainDeferred", t0);
+    if (stallDebug) acc(&accDeferred, t0);
     if (stopped_) break;
     armTimerFd();
     int timeoutMs = deferred_.empty() ? 1000 : 0;
+    TimePoint te = Clock::now();
     int n = epoll_wait(epfd_, events.data(), (int)events.size(), timeoutMs);
+    if (stallDebug) acc(&accEpoll, te);
     if (n < 0) {
       if (errno == EINTR) continue;
       LOG_ERROR("epoll_wait: %s", strerror(errno));
       break;
     }
+    TimePoint tf = Clock::now();
     for (int i = 0; i < n && !stopped_; i++) {
       int fd = events[i].data.fd;
       auto it = fdCallbacks_.find(fd);
@@ -338,9 +350,26 @@ void Loop::run() {
         probe("fdCallback", tc, fd);
       }
     }
+    if (stallDebug) acc(&accFds, tf);
     TimePoint tt = Clock::now();
     fireDueTimers();
     probe("fireDueTimers", tt);
+    if (stallDebug) {
+      acc(&accTimers, tt);
+      iters++;
+      if (Clock::now() - lastReport > std::chrono::seconds(5)) {
+        double secs = std::chrono::duration<double>(Clock::now() -
+                                                    lastReport).count();
+        fprintf(stderr,
+                "loop phases: %.0f it/s deferred=%.0fms epoll=%.0fms "
+                "fds=%.0fms timers=%.0fms (per %.1fs)\n",
+                iters / secs, accDeferred, accEpoll, accFds, accTimers,
+                secs);
+        accDeferred = accEpoll = accFds = accTimers = 0;
+        iters = 0;
+        lastReport = Clock::now();
+      }
+    }
   }
 }
 

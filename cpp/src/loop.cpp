@@ -10,6 +10,7 @@
 
 #include <cstdio>
 #include <cstring>
+#include <algorithm>
 #include <set>
 #include <stdexcept>
 
@@ -316,6 +317,7 @@ void Loop::run() {
   std::vector<struct epoll_event> events(64);
   // phase accounting (stallDebug): 5 s summaries of where loop time goes
   double accDeferred = 0, accEpoll = 0, accFds = 0, accTimers = 0;
+  std::map<int, std::pair<double, uint64_t>> accPerFd;  // fd -> (ms, count)
   uint64_t iters = 0;
   TimePoint lastReport = Clock::now();
   auto acc = [](double* slot, TimePoint t0) {
@@ -348,6 +350,12 @@ void Loop::run() {
         TimePoint tc = Clock::now();
         cb(events[i].events);
         probe("fdCallback", tc, fd);
+        if (stallDebug) {
+          auto& slot = accPerFd[fd];
+          slot.first += std::chrono::duration<double, std::milli>(
+                            Clock::now() - tc).count();
+          slot.second++;
+        }
       }
     }
     if (stallDebug) acc(&accFds, tf);
@@ -365,6 +373,15 @@ void Loop::run() {
                 "fds=%.0fms timers=%.0fms (per %.1fs)\n",
                 iters / secs, accDeferred, accEpoll, accFds, accTimers,
                 secs);
+        std::vector<std::pair<double, std::pair<int, uint64_t>>> top;
+        for (auto& kv : accPerFd)
+          top.push_back({kv.second.first, {kv.first, kv.second.second}});
+        std::sort(top.rbegin(), top.rend());
+        for (size_t k = 0; k < top.size() && k < 4; k++)
+          fprintf(stderr, "  top fd %d: %.0fms over %llu events\n",
+                  top[k].second.first, top[k].first,
+                  (unsigned long long)top[k].second.second);
+        accPerFd.clear();
         accDeferred = accEpoll = accFds = accTimers = 0;
         iters = 0;
         lastReport = Clock::now();

@@ -68,6 +68,7 @@ print(json.dumps({
                                   max(1, len(rtts)), 1),
     "rtt_mean_exec_phase": round(sum(r[2] for r in rtts) /
                                  max(1, len(rtts)), 1),
-    "loop_phases": re.findall(r"loop phases: .*", log)[-4:],
+    "loop_phases": re.findall(r"loop phases: .*", log)[-2:],
+    "top_fds": re.findall(r"top fd .*", log)[-8:],
 }), flush=True)
 d.cleanup()

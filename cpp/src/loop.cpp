@@ -289,7 +289,34 @@ void Loop::drainPosted() {
       }
       more = !posted_.empty();
     }
-    for (auto& fn : batch) fn();
+    static const bool itemDebug = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
+    if (itemDebug) {
+      static double accMs = 0;
+      static uint64_t nItems = 0, nSlow = 0;
+      static TimePoint lastDump = Clock::now();
+      for (auto& fn : batch) {
+        TimePoint ti = Clock::now();
+        fn();
+        double ms = std::chrono::duration<double, std::milli>(
+                        Clock::now() - ti).count();
+        accMs += ms;
+        nItems++;
+        if (ms > 0.2) nSlow++;
+      }
+      if (Clock::now() - lastDump > std::chrono::seconds(5)) {
+        fprintf(stderr,
+                "posted items: %llu in 5s, total=%.0fms avg=%.1fus "
+                "slow(>200us)=%llu\n",
+                (unsigned long long)nItems, accMs,
+                nItems ? accMs * 1000.0 / nItems : 0.0,
+                (unsigned long long)nSlow);
+        accMs = 0;
+        nItems = nSlow = 0;
+        lastDump = Clock::now();
+      }
+    } else {
+      for (auto& fn : batch) fn();
+    }
     processed += batch.size();
     if (batch.empty()) break;
   }

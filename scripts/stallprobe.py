@@ -69,6 +69,7 @@ print(json.dumps({
     "rtt_mean_exec_phase": round(sum(r[2] for r in rtts) /
                                  max(1, len(rtts)), 1),
     "loop_phases": re.findall(r"loop phases: .*", log)[-2:],
-    "top_fds": re.findall(r"top fd .*", log)[-8:],
+    "top_fds": re.findall(r"top fd .*", log)[-4:],
+    "posted_items": re.findall(r"posted items: .*", log)[-3:],
 }), flush=True)
 d.cleanup()

@@ -26,6 +26,12 @@ using Clock = std::chrono::steady_clock;
 using TimePoint = Clock::time_point;
 using Ns = std::chrono::nanoseconds;
 
+// Debug helpers (CPILOT_LOOP_DEBUG): wrap a posted callback so its
+// execution cost is accumulated under `tag`; the loop's 5 s phase
+// report prints per-tag totals. No-ops (returns fn) when debug is off.
+std::function<void()> timedItem(const char* tag, std::function<void()> fn);
+void dumpItemTags();
+
 // Reset the calling thread to default scheduling (SCHED_OTHER, nice 0).
 // The daemon's reactor thread runs elevated (SCHED_RR or negative nice);
 // background threads and child processes must not inherit that.

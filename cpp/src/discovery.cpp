@@ -274,7 +274,7 @@ void ConsulBackend::serviceRegister(
     std::string err = res.ok ? ("status " + std::to_string(res.status) + ": " +
                                 res.body)
                              : res.error;
-    loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
+    loop->post(timedItem("consul", [cb, ok, err] { cb(ok, ok ? "" : err); }));
   });
   if (!accepted)
     loop_->post([cb] { cb(false, "request coalesced or queue full"); });
@@ -308,7 +308,7 @@ void ConsulBackend::updateTTL(const std::string& checkID,
     std::string err =
         res.ok ? ("status " + std::to_string(res.status) + ": " + res.body)
                : res.error;
-    loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
+    loop->post(timedItem("consul", [cb, ok, err] { cb(ok, ok ? "" : err); }));
   });
   if (!accepted) {
     // a heartbeat for this check is already queued; nothing to report
@@ -330,7 +330,7 @@ void ConsulBackend::serviceDeregister(const std::string& id, DoneCb cb) {
     bool ok = res.ok && res.status == 200;
     std::string err =
         res.ok ? ("status " + std::to_string(res.status)) : res.error;
-    loop->post([cb, ok, err] { cb(ok, ok ? "" : err); });
+    loop->post(timedItem("consul", [cb, ok, err] { cb(ok, ok ? "" : err); }));
   });
   if (!accepted) loop_->post([cb] { cb(false, "queue full"); });
 }
@@ -383,9 +383,10 @@ void ConsulBackend::healthService(const std::string& name,
     bool ok = res.ok && res.status == 200;
     std::vector<ServiceEntry> entries;
     if (ok) entries = parseHealthEntries(res.body, &ok);
-    loop->post([cb, ok, entries = std::move(entries)]() mutable {
-      cb(ok, std::move(entries));
-    });
+    loop->post(timedItem("consul",
+                         [cb, ok, entries = std::move(entries)]() mutable {
+                           cb(ok, std::move(entries));
+                         }));
   });
   if (!accepted) {
     loop_->post([cb] { cb(false, {}); });

@@ -38,6 +38,40 @@ std::set<Loop*>& liveLoops() {
 }
 }  // namespace
 
+namespace {
+struct ItemTagStats {
+  double ms = 0;
+  uint64_t count = 0;
+};
+std::map<std::string, ItemTagStats>& itemTags() {
+  static std::map<std::string, ItemTagStats> tags;  // loop thread only
+  return tags;
+}
+}  // namespace
+
+std::function<void()> timedItem(const char* tag, std::function<void()> fn) {
+  static const bool on = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
+  if (!on) return fn;
+  std::string t(tag);
+  return [t, fn = std::move(fn)] {
+    TimePoint t0 = Clock::now();
+    fn();
+    auto& slot = itemTags()[t];
+    slot.ms +=
+        std::chrono::duration<double, std::milli>(Clock::now() - t0).count();
+    slot.count++;
+  };
+}
+
+void dumpItemTags() {
+  for (auto& kv : itemTags())
+    fprintf(stderr, "  item tag %s: %.0fms over %llu (avg %.1fus)\n",
+            kv.first.c_str(), kv.second.ms,
+            (unsigned long long)kv.second.count,
+            kv.second.count ? kv.second.ms * 1000 / kv.second.count : 0.0);
+  itemTags().clear();
+}
+
 void Loop::postIfLive(Loop* loop, std::function<void()> fn) {
   std::lock_guard<std::mutex> l(liveLoopsMu());
   if (liveLoops().count(loop)) loop->post(std::move(fn));
@@ -408,6 +442,7 @@ void Loop::run() {
           fprintf(stderr, "  top fd %d: %.0fms over %llu events\n",
                   top[k].second.first, top[k].first,
                   (unsigned long long)top[k].second.second);
+        dumpItemTags();
         accPerFd.clear();
         accDeferred = accEpoll = accFds = accTimers = 0;
         iters = 0;

@@ -336,7 +336,8 @@ void Spawner::handleMessage(size_t idx, const char* data, size_t len) {
     SpawnCb cb = std::move(p.cb);
     pid_t pid = r.pid;
     int err = r.err;
-    Loop::postIfLive(p.loop, [cb, pid, err] { cb(pid, err); });
+    Loop::postIfLive(p.loop,
+                     timedItem("spawncb", [cb, pid, err] { cb(pid, err); }));
   } else if (type == kChildExit && len >= sizeof(ExitNotice)) {
     ExitNotice note;
     memcpy(&note, data, sizeof(note));
@@ -353,8 +354,9 @@ void Spawner::handleMessage(size_t idx, const char* data, size_t len) {
     pidLoops_.erase(it);
     pid_t pid = note.pid;
     int status = note.status;
-    Loop::postIfLive(loop,
-                     [loop, pid, status] { loop->notifyChildExit(pid, status); });
+    Loop::postIfLive(loop, timedItem("exit", [loop, pid, status] {
+                       loop->notifyChildExit(pid, status);
+                     }));
   }
 }
 

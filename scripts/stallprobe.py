@@ -70,6 +70,7 @@ print(json.dumps({
                                  max(1, len(rtts)), 1),
     "loop_phases": re.findall(r"loop phases: .*", log)[-2:],
     "top_fds": re.findall(r"top fd .*", log)[-4:],
-    "posted_items": re.findall(r"posted items: .*", log)[-3:],
+    "posted_items": re.findall(r"posted items: .*", log)[-2:],
+    "item_tags": re.findall(r"item tag .*", log)[-6:],
 }), flush=True)
 d.cleanup()

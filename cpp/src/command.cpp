@@ -294,6 +294,19 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   lap("spawnerEnqueue");
 }
 
+namespace {
+// CPILOT_LOOP_DEBUG: accumulated per-span cost inside the exit path
+struct ExitSpans {
+  double logdrain = 0, overlay = 0, publish = 0, respawn = 0;
+  uint64_t n = 0;
+  TimePoint last = Clock::now();
+};
+ExitSpans& exitSpans() {
+  static ExitSpans spans;
+  return spans;
+}
+}  // namespace
+
 void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
   LOG_DEBUG("%s.Run end", name_.c_str());
   static const bool rttDebug = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
@@ -310,6 +323,13 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
               name_.c_str(), total, total - execPhase, execPhase);
     }
   }
+  TimePoint tSpan = Clock::now();
+  auto lap = [&](double* slot) {
+    if (!rttDebug) return;
+    auto now = Clock::now();
+    *slot += std::chrono::duration<double, std::milli>(now - tSpan).count();
+    tSpan = now;
+  };
   if (timeoutTimer_) {
     loop.cancelTimer(timeoutTimer_);
     timeoutTimer_ = 0;
@@ -334,7 +354,9 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
     logFd_ = -1;
   }
 
+  lap(&exitSpans().logdrain);
   if (!pidEnvName_.empty()) overlayErase(pidEnvName_);
+  lap(&exitSpans().overlay);
 
   running_ = false;
   pid_ = -1;
@@ -352,9 +374,23 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
                       name_ + ": exit status " + std::to_string(code)});
   }
 
+  lap(&exitSpans().publish);
   if (pendingRun_) {
     pendingRun_ = false;
     spawn(loop, bus);
+  }
+  lap(&exitSpans().respawn);
+  if (rttDebug) {
+    auto& sp = exitSpans();
+    sp.n++;
+    if (Clock::now() - sp.last > std::chrono::seconds(5)) {
+      fprintf(stderr,
+              "exit spans: n=%llu logdrain=%.0fms overlay=%.0fms "
+              "publish=%.0fms respawn=%.0fms\n",
+              (unsigned long long)sp.n, sp.logdrain, sp.overlay, sp.publish,
+              sp.respawn);
+      sp = ExitSpans{};
+    }
   }
 }
 

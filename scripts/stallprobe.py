@@ -71,6 +71,7 @@ print(json.dumps({
     "loop_phases": re.findall(r"loop phases: .*", log)[-2:],
     "top_fds": re.findall(r"top fd .*", log)[-4:],
     "posted_items": re.findall(r"posted items: .*", log)[-2:],
-    "item_tags": re.findall(r"item tag .*", log)[-6:],
+    "item_tags": re.findall(r"item tag .*", log)[-3:],
+    "exit_spans": re.findall(r"exit spans: .*", log)[-2:],
 }), flush=True)
 d.cleanup()

@@ -203,7 +203,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
     }
   }
 
-  lap("pipe2");
+  lap("pipe2", &accPipe);
   running_ = true;
   pid_ = -1;
   pendingSignal_ = 0;
@@ -213,7 +213,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   // snapshot env (+ the PID-var overlay) on this (the loop) thread; the
   // spawner must never read the live environ concurrently with setenv
   auto baseEnv = baseEnvSnapshot();
-  lap("envSnapshot");
+  lap("envSnapshot", &accEnv);
   // the spawner pool does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
   Spawner::global().spawn(
@@ -291,7 +291,18 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
           onExit(*loop_, bus, status);
         });
       });
-  lap("spawnerEnqueue");
+  lap("spawnerEnqueue", &accEnq);
+  if (spdbg) {
+    accN++;
+    if (Clock::now() - accLast > std::chrono::seconds(5)) {
+      fprintf(stderr,
+              "spawn spans: n=%llu pipe2=%.0fms env=%.0fms enqueue=%.0fms\n",
+              (unsigned long long)accN, accPipe, accEnv, accEnq);
+      accPipe = accEnv = accEnq = 0;
+      accN = 0;
+      accLast = Clock::now();
+    }
+  }
 }
 
 namespace {

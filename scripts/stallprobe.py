@@ -73,5 +73,6 @@ print(json.dumps({
     "posted_items": re.findall(r"posted items: .*", log)[-2:],
     "item_tags": re.findall(r"item tag .*", log)[-3:],
     "exit_spans": re.findall(r"exit spans: .*", log)[-2:],
+    "spawn_spans": re.findall(r"spawn spans: .*", log)[-2:],
 }), flush=True)
 d.cleanup()

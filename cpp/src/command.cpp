@@ -182,14 +182,18 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   loop_ = &loop;
   LOG_DEBUG("%s.Run start", name_.c_str());
   static const bool spdbg = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
+  static double accPipe = 0, accEnv = 0, accEnq = 0;
+  static uint64_t accN = 0;
+  static TimePoint accLast = Clock::now();
   auto tstart = Clock::now();
-  auto lap = [&](const char* what) {
+  auto lap = [&](const char* what, double* slot) {
     if (!spdbg) return;
-    auto ms = std::chrono::duration<double, std::milli>(Clock::now() -
-                                                        tstart).count();
+    auto now = Clock::now();
+    auto ms = std::chrono::duration<double, std::milli>(now - tstart).count();
+    *slot += ms;
     if (ms > 5.0)
       fprintf(stderr, "spawn-step stall: %s after %.1f ms\n", what, ms);
-    tstart = Clock::now();
+    tstart = now;
   };
 
   int pipefds[2] = {-1, -1};

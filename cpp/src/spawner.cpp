@@ -71,9 +71,9 @@ Spawner::Spawner() {
   if (pipe2(wakeFds_, O_NONBLOCK | O_CLOEXEC) != 0)
     LOG_ERROR("spawner: pipe2 failed: %s", strerror(errno));
   int n = defaultHelperCount();
-  helpers_.resize(n);
+  for (int i = 0; i < n; i++) helpers_.push_back(std::make_unique<Helper>());
   for (auto& h : helpers_) {
-    if (!launchHelper(&h)) {
+    if (!launchHelper(h.get())) {
       logging::logf(logging::Level::Fatal,
                     "spawner: cannot launch spawn helper '%s': %s",
                     helperPath_.c_str(), strerror(errno));
@@ -128,6 +128,41 @@ size_t Spawner::backlog() {
   return pending_.size();
 }
 
+// Drain a helper's overflow: pop a batch under mu_, send outside it
+// (only the reader calls this, so within-overflow order is preserved;
+// a concurrent direct send from the loop may interleave, which is fine
+// — spawn requests of distinct commands are order-independent).
+void Spawner::flushOverflow(size_t idx) {
+  Helper& h = *helpers_[idx];
+  while (true) {
+    std::deque<std::pair<std::vector<char>, int>> batch;
+    {
+      std::lock_guard<std::mutex> l(mu_);
+      if (h.dead || h.overflow.empty()) return;
+      for (int i = 0; i < 16 && !h.overflow.empty(); i++) {
+        batch.push_back(std::move(h.overflow.front()));
+        h.overflow.pop_front();
+      }
+    }
+    size_t sent = 0;
+    {
+      std::lock_guard<std::mutex> sl(h.sendMu);
+      for (auto& [buf, fd] : batch) {
+        if (!sendRequest(h, buf, fd)) break;
+        if (fd >= 0) close(fd);
+        sent++;
+      }
+    }
+    if (sent < batch.size()) {
+      // kernel buffer full again: put the unsent tail back at the front
+      std::lock_guard<std::mutex> l(mu_);
+      for (size_t i = batch.size(); i > sent; i--)
+        h.overflow.push_front(std::move(batch[i - 1]));
+      return;
+    }
+  }
+}
+
 bool Spawner::overloaded() {
   // ~16 outstanding per helper ≈ a few ms of queue at measured spawn
   // cost; beyond that a new check would only inflate round trips
@@ -178,38 +213,59 @@ void Spawner::spawn(Loop& loop, std::string execPath,
   for (auto& e : extraEnv) append(e);
 
   TimePoint tLock = Clock::now();
-  std::lock_guard<std::mutex> l(mu_);
-  if (spawnDebug()) {
-    auto ms = std::chrono::duration<double, std::milli>(Clock::now() -
-                                                        tLock).count();
-    if (ms > 5.0) fprintf(stderr, "spawner mu_ wait %.1f ms\n", ms);
-  }
-  uint32_t reqId = nextReqId_++;
-  hdr.reqId = reqId;
-  memcpy(buf.data(), &hdr, sizeof(hdr));
-
-  // round-robin over live helpers
   size_t idx = 0;
-  bool found = false;
-  for (size_t i = 0; i < helpers_.size(); i++) {
-    idx = (nextHelper_ + i) % helpers_.size();
-    if (!helpers_[idx].dead) {
-      found = true;
-      break;
+  uint32_t reqId = 0;
+  bool queued = false;
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    if (spawnDebug()) {
+      auto ms = std::chrono::duration<double, std::milli>(Clock::now() -
+                                                          tLock).count();
+      if (ms > 5.0) fprintf(stderr, "spawner mu_ wait %.1f ms\n", ms);
+    }
+    reqId = nextReqId_++;
+    hdr.reqId = reqId;
+    memcpy(buf.data(), &hdr, sizeof(hdr));
+
+    // round-robin over live helpers
+    bool found = false;
+    for (size_t i = 0; i < helpers_.size(); i++) {
+      idx = (nextHelper_ + i) % helpers_.size();
+      if (!helpers_[idx]->dead) {
+        found = true;
+        break;
+      }
+    }
+    nextHelper_ = (idx + 1) % helpers_.size();
+    if (!found) {
+      if (stdioFd >= 0) close(stdioFd);
+      loop.post([cb] { cb(-1, ECHILD); });
+      return;
+    }
+    pending_[reqId] = Pending{&loop, std::move(cb), idx};
+    if (!helpers_[idx]->overflow.empty()) {
+      // backlog exists: append behind it (rough fairness)
+      helpers_[idx]->overflow.emplace_back(std::move(buf), stdioFd);
+      queued = true;
     }
   }
-  nextHelper_ = (idx + 1) % helpers_.size();
-  if (!found) {
-    if (stdioFd >= 0) close(stdioFd);
-    loop.post([cb] { cb(-1, ECHILD); });
-    return;
-  }
-  Helper& h = helpers_[idx];
-  pending_[reqId] = Pending{&loop, std::move(cb), idx};
   if (spawnDebug())
     fprintf(stderr, "[spawner] req %u -> helper %zu\n", reqId, idx);
-  if (!h.overflow.empty() || !sendRequest(h, buf, stdioFd)) {
-    // order-preserving overflow; the fd stays open until actually sent
+  if (queued) {
+    wakeReader();
+    return;
+  }
+  // send outside mu_: only this helper's sendMu serializes the syscall,
+  // so the reader's overflow flushes never block spawn admission
+  Helper& h = *helpers_[idx];
+  bool sent;
+  {
+    std::lock_guard<std::mutex> sl(h.sendMu);
+    sent = sendRequest(h, buf, stdioFd);
+  }
+  if (!sent) {
+    // the fd stays open until actually sent
+    std::lock_guard<std::mutex> l(mu_);
     h.overflow.emplace_back(std::move(buf), stdioFd);
     wakeReader();
     return;
@@ -255,9 +311,9 @@ void Spawner::readerMain() {
     {
       std::lock_guard<std::mutex> l(mu_);
       for (auto& h : helpers_) {
-        short ev = h.dead ? 0 : POLLIN;
-        if (!h.dead && !h.overflow.empty()) ev |= POLLOUT;
-        fds.push_back({h.sock, ev, 0});
+        short ev = h->dead ? 0 : POLLIN;
+        if (!h->dead && !h->overflow.empty()) ev |= POLLOUT;
+        fds.push_back({h->sock, ev, 0});
       }
     }
     fds.push_back({wakeFds_[0], POLLIN, 0});
@@ -278,16 +334,7 @@ void Spawner::readerMain() {
       }
     }
     for (size_t i = 0; i + 1 < fds.size(); i++) {
-      if (fds[i].revents & POLLOUT) {
-        std::lock_guard<std::mutex> l(mu_);
-        Helper& h = helpers_[i];
-        while (!h.overflow.empty()) {
-          auto& [obuf, ofd] = h.overflow.front();
-          if (!sendRequest(h, obuf, ofd)) break;
-          if (ofd >= 0) close(ofd);
-          h.overflow.pop_front();
-        }
-      }
+      if (fds[i].revents & POLLOUT) flushOverflow(i);
       if (fds[i].revents & (POLLIN | POLLHUP | POLLERR)) {
         while (true) {
           ssize_t n = recv(fds[i].fd, buf.data(), buf.size(), MSG_DONTWAIT);
@@ -364,7 +411,7 @@ void Spawner::helperDied(size_t idx) {
   std::vector<Pending> lost;
   {
     std::lock_guard<std::mutex> l(mu_);
-    Helper& h = helpers_[idx];
+    Helper& h = *helpers_[idx];
     if (h.dead) return;
     h.dead = true;
     LOG_ERROR("spawner: helper %d (pid %d) died; respawning",

@@ -251,8 +251,28 @@ void Spawner::spawn(Loop& loop, std::string execPath,
   }
   if (spawnDebug())
     fprintf(stderr, "[spawner] req %u -> helper %zu\n", reqId, idx);
+  static double accQueueMs = 0, accLockMs = 0, accSendMs = 0;
+  static uint64_t nQueued = 0, nDirect = 0;
+  static TimePoint accLast2 = Clock::now();
+  static const bool dbg2 = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
+  if (dbg2 && Clock::now() - accLast2 > std::chrono::seconds(5)) {
+    fprintf(stderr,
+            "spawn branches: queued=%llu direct=%llu queueMs=%.0f "
+            "lockMs=%.0f sendMs=%.0f\n",
+            (unsigned long long)nQueued, (unsigned long long)nDirect,
+            accQueueMs, accLockMs, accSendMs);
+    nQueued = nDirect = 0;
+    accQueueMs = accLockMs = accSendMs = 0;
+    accLast2 = Clock::now();
+  }
   if (queued) {
+    TimePoint tw = Clock::now();
     wakeReader();
+    if (dbg2) {
+      nQueued++;
+      accQueueMs += std::chrono::duration<double, std::milli>(
+                        Clock::now() - tw).count();
+    }
     return;
   }
   // send outside mu_: only this helper's sendMu serializes the syscall,
@@ -260,8 +280,16 @@ void Spawner::spawn(Loop& loop, std::string execPath,
   Helper& h = *helpers_[idx];
   bool sent;
   {
+    TimePoint tl = Clock::now();
     std::lock_guard<std::mutex> sl(h.sendMu);
+    TimePoint ts = Clock::now();
     sent = sendRequest(h, buf, stdioFd);
+    if (dbg2) {
+      nDirect++;
+      accLockMs += std::chrono::duration<double, std::milli>(ts - tl).count();
+      accSendMs += std::chrono::duration<double, std::milli>(
+                       Clock::now() - ts).count();
+    }
   }
   if (!sent) {
     // the fd stays open until actually sent

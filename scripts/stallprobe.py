@@ -74,5 +74,6 @@ print(json.dumps({
     "item_tags": re.findall(r"item tag .*", log)[-3:],
     "exit_spans": re.findall(r"exit spans: .*", log)[-2:],
     "spawn_spans": re.findall(r"spawn spans: .*", log)[-2:],
+    "spawn_branches": re.findall(r"spawn branches: .*", log)[-2:],
 }), flush=True)
 d.cleanup()

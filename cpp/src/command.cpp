@@ -220,9 +220,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   lap("envSnapshot", &accEnv);
   // the spawner pool does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
-  Spawner::global().spawn(
-      loop, exec_, args_, std::move(baseEnv), overlayExtras(),
-      raw_ ? -1 : pipefds[1],
+  Spawner::SpawnCb compCb =
       [this, self, bus, readFd](pid_t pid, int err) {
         if (pid < 0) {
           LOG_ERROR("unable to start %s: %s", name_.c_str(), strerror(err));

@@ -188,6 +188,17 @@ void Spawner::spawn(Loop& loop, std::string execPath,
   hdr.envc = (uint32_t)((baseEnv ? baseEnv->size() : 0) + extraEnv.size());
   hdr.wantStdio = stdioFd >= 0 ? 1 : 0;
 
+  static double accSer = 0, accMu = 0, accPost = 0;
+  static uint64_t accCnt = 0;
+  static TimePoint accT = Clock::now();
+  static const bool dbg3 = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
+  TimePoint tphase = Clock::now();
+  auto lap3 = [&](double* slot) {
+    if (!dbg3) return;
+    auto now = Clock::now();
+    *slot += std::chrono::duration<double, std::milli>(now - tphase).count();
+    tphase = now;
+  };
   size_t size = sizeof(hdr) + execPath.size() + 1;
   for (auto& a : args) size += a.size() + 1;
   if (baseEnv)
@@ -212,6 +223,7 @@ void Spawner::spawn(Loop& loop, std::string execPath,
     for (auto& e : *baseEnv) append(e);
   for (auto& e : extraEnv) append(e);
 
+  lap3(&accSer);
   TimePoint tLock = Clock::now();
   size_t idx = 0;
   uint32_t reqId = 0;
@@ -249,6 +261,7 @@ void Spawner::spawn(Loop& loop, std::string execPath,
       queued = true;
     }
   }
+  lap3(&accMu);
   if (spawnDebug())
     fprintf(stderr, "[spawner] req %u -> helper %zu\n", reqId, idx);
   static double accQueueMs = 0, accLockMs = 0, accSendMs = 0;
@@ -299,6 +312,18 @@ void Spawner::spawn(Loop& loop, std::string execPath,
     return;
   }
   if (stdioFd >= 0) close(stdioFd);
+  lap3(&accPost);
+  if (dbg3) {
+    accCnt++;
+    if (Clock::now() - accT > std::chrono::seconds(5)) {
+      fprintf(stderr, "spawner inner: n=%llu serialize=%.0fms mu=%.0fms "
+              "post=%.0fms\n",
+              (unsigned long long)accCnt, accSer, accMu, accPost);
+      accSer = accMu = accPost = 0;
+      accCnt = 0;
+      accT = Clock::now();
+    }
+  }
 }
 
 // Send one request datagram (+fd). Returns false on EAGAIN (caller

@@ -74,6 +74,7 @@ print(json.dumps({
     "item_tags": re.findall(r"item tag .*", log)[-3:],
     "exit_spans": re.findall(r"exit spans: .*", log)[-2:],
     "spawn_spans": re.findall(r"spawn spans: .*", log)[-2:],
-    "spawn_branches": re.findall(r"spawn branches: .*", log)[-2:],
+    "spawn_branches": re.findall(r"spawn branches: .*", log)[-1:],
+    "spawner_inner": re.findall(r"spawner inner: .*", log)[-2:],
 }), flush=True)
 d.cleanup()

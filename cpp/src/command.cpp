@@ -292,7 +292,11 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
         loop_->watchChild(pid, [this, self, bus](int status) {
           onExit(*loop_, bus, status);
         });
-      });
+      };
+  lap("lambdaBuild", &accEnv);
+  Spawner::global().spawn(loop, exec_, args_, std::move(baseEnv),
+                          overlayExtras(), raw_ ? -1 : pipefds[1],
+                          std::move(compCb));
   lap("spawnerEnqueue", &accEnq);
   if (spdbg) {
     accN++;

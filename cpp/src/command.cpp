@@ -294,6 +294,7 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
         });
       };
   lap("lambdaBuild", &accEnv);
+  if (spdbg) g_spawnCallT0 = Clock::now();
   Spawner::global().spawn(loop, exec_, args_, std::move(baseEnv),
                           overlayExtras(), raw_ ? -1 : pipefds[1],
                           std::move(compCb));

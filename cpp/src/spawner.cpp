@@ -59,6 +59,8 @@ bool spawnDebug() {
 
 }  // namespace
 
+thread_local Clock::time_point g_spawnCallT0{};
+
 Spawner& Spawner::global() {
   // intentionally leaked: the reader thread and helper processes live
   // for the process lifetime
@@ -188,11 +190,16 @@ void Spawner::spawn(Loop& loop, std::string execPath,
   hdr.envc = (uint32_t)((baseEnv ? baseEnv->size() : 0) + extraEnv.size());
   hdr.wantStdio = stdioFd >= 0 ? 1 : 0;
 
-  static double accSer = 0, accMu = 0, accPost = 0;
+  static double accSer = 0, accMu = 0, accPost = 0, accGap = 0;
   static uint64_t accCnt = 0;
   static TimePoint accT = Clock::now();
   static const bool dbg3 = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
   TimePoint tphase = Clock::now();
+  if (dbg3 && g_spawnCallT0.time_since_epoch().count() != 0) {
+    accGap += std::chrono::duration<double, std::milli>(
+                  tphase - g_spawnCallT0).count();
+    g_spawnCallT0 = Clock::time_point{};
+  }
   auto lap3 = [&](double* slot) {
     if (!dbg3) return;
     auto now = Clock::now();
@@ -316,10 +323,10 @@ void Spawner::spawn(Loop& loop, std::string execPath,
   if (dbg3) {
     accCnt++;
     if (Clock::now() - accT > std::chrono::seconds(5)) {
-      fprintf(stderr, "spawner inner: n=%llu serialize=%.0fms mu=%.0fms "
-              "post=%.0fms\n",
-              (unsigned long long)accCnt, accSer, accMu, accPost);
-      accSer = accMu = accPost = 0;
+      fprintf(stderr, "spawner inner: n=%llu gap=%.0fms serialize=%.0fms "
+              "mu=%.0fms post=%.0fms\n",
+              (unsigned long long)accCnt, accGap, accSer, accMu, accPost);
+      accSer = accMu = accPost = accGap = 0;
       accCnt = 0;
       accT = Clock::now();
     }

@@ -50,9 +50,14 @@ class Spawner {
   // baseEnv is an immutable shared snapshot of the environment taken ON
   // THE LOOP THREAD; extraEnv are per-spawn overlay entries ("K=V")
   // appended after it.
-  void spawn(Loop& loop, std::string execPath, std::vector<std::string> args,
+  // extraEnv is serialized synchronously before spawn() returns, so a
+  // reference to a loop-thread-owned vector is safe and avoids an
+  // O(overlay) copy per spawn
+  void spawn(Loop& loop, const std::string& execPath,
+             const std::vector<std::string>& args,
              std::shared_ptr<const std::vector<std::string>> baseEnv,
-             std::vector<std::string> extraEnv, int stdioFd, SpawnCb cb);
+             const std::vector<std::string>& extraEnv, int stdioFd,
+             SpawnCb cb);
 
   int helperCount() const { return (int)helpers_.size(); }
 

@@ -177,10 +177,10 @@ void Spawner::wakeReader() {
   (void)unused;
 }
 
-void Spawner::spawn(Loop& loop, std::string execPath,
-                    std::vector<std::string> args,
+void Spawner::spawn(Loop& loop, const std::string& execPath,
+                    const std::vector<std::string>& args,
                     std::shared_ptr<const std::vector<std::string>> baseEnv,
-                    std::vector<std::string> extraEnv, int stdioFd,
+                    const std::vector<std::string>& extraEnv, int stdioFd,
                     SpawnCb cb) {
   // serialize: header + argv strings + env strings, NUL-terminated
   RequestHeader hdr;

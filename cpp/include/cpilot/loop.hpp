@@ -79,10 +79,14 @@ class Loop {
   // spawner posts these). Same claimed/unclaimed logic as reapChildren.
   void notifyChildExit(pid_t pid, int status);
 
-  // Liveness-checked cross-thread post: delivers fn to `loop` only if it
-  // still exists (spawn-helper exit notices can outlive the Loop that
-  // requested the spawn, e.g. unit tests destroying per-test loops).
-  static void postIfLive(Loop* loop, std::function<void()> fn);
+  // Liveness-checked cross-thread post: delivers fn only if `loop` still
+  // exists AND is the same incarnation (`id`): spawn-helper replies and
+  // exit notices can outlive the Loop that requested the spawn, and a
+  // NEW loop can be constructed at the SAME address (stack-allocated
+  // per-test loops reuse frames), so pointer identity alone would
+  // deliver a dead generation's callbacks into the wrong loop.
+  static void postIfLive(Loop* loop, uint64_t id, std::function<void()> fn);
+  uint64_t id() const { return id_; }
 
   void run();   // until stop()
   void stop();
@@ -109,6 +113,7 @@ class Loop {
   void drainDeferred();
   void drainPosted();
 
+  uint64_t id_ = 0;  // unique per Loop incarnation
   int epfd_ = -1;
   int timerfd_ = -1;
   int wakeupFds_[2] = {-1, -1};  // pipe for post()

@@ -87,6 +87,7 @@ class Spawner {
   };
   struct Pending {
     Loop* loop;
+    uint64_t loopId;
     SpawnCb cb;
     size_t helperIdx;
   };
@@ -106,8 +107,8 @@ class Spawner {
   std::mutex mu_;  // guards pending_, overflow queues, helper respawn
   std::map<uint32_t, Pending> pending_;
   uint32_t nextReqId_ = 1;
-  // pid -> loop for exit notices (reader thread only)
-  std::map<pid_t, Loop*> pidLoops_;
+  // pid -> (loop, incarnation id) for exit notices (reader thread only)
+  std::map<pid_t, std::pair<Loop*, uint64_t>> pidLoops_;
 
   int wakeFds_[2] = {-1, -1};
   std::thread reader_;

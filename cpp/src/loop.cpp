@@ -32,9 +32,13 @@ std::mutex& liveLoopsMu() {
   static std::mutex mu;
   return mu;
 }
-std::set<Loop*>& liveLoops() {
-  static std::set<Loop*> loops;
+std::map<Loop*, uint64_t>& liveLoops() {
+  static std::map<Loop*, uint64_t> loops;
   return loops;
+}
+uint64_t nextLoopId() {
+  static uint64_t next = 1;
+  return next++;  // guarded by liveLoopsMu
 }
 }  // namespace
 
@@ -72,15 +76,17 @@ void dumpItemTags() {
   itemTags().clear();
 }
 
-void Loop::postIfLive(Loop* loop, std::function<void()> fn) {
+void Loop::postIfLive(Loop* loop, uint64_t id, std::function<void()> fn) {
   std::lock_guard<std::mutex> l(liveLoopsMu());
-  if (liveLoops().count(loop)) loop->post(std::move(fn));
+  auto it = liveLoops().find(loop);
+  if (it != liveLoops().end() && it->second == id) loop->post(std::move(fn));
 }
 
 Loop::Loop() {
   {
     std::lock_guard<std::mutex> l(liveLoopsMu());
-    liveLoops().insert(this);
+    id_ = nextLoopId();
+    liveLoops()[this] = id_;
   }
   epfd_ = epoll_create1(EPOLL_CLOEXEC);
   if (epfd_ < 0) throw std::runtime_error("epoll_create1 failed");

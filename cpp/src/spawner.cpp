@@ -261,7 +261,7 @@ void Spawner::spawn(Loop& loop, const std::string& execPath,
       loop.post([cb] { cb(-1, ECHILD); });
       return;
     }
-    pending_[reqId] = Pending{&loop, std::move(cb), idx};
+    pending_[reqId] = Pending{&loop, loop.id(), std::move(cb), idx};
     if (!helpers_[idx]->overflow.empty()) {
       // backlog exists: append behind it (rough fairness)
       helpers_[idx]->overflow.emplace_back(std::move(buf), stdioFd);
@@ -439,11 +439,11 @@ void Spawner::handleMessage(size_t idx, const char* data, size_t len) {
     }
     if (spawnDebug())
       fprintf(stderr, "[spawner] reply req %u pid %d\n", r.reqId, r.pid);
-    if (r.pid > 0) pidLoops_[r.pid] = p.loop;
+    if (r.pid > 0) pidLoops_[r.pid] = {p.loop, p.loopId};
     SpawnCb cb = std::move(p.cb);
     pid_t pid = r.pid;
     int err = r.err;
-    Loop::postIfLive(p.loop,
+    Loop::postIfLive(p.loop, p.loopId,
                      timedItem("spawncb", [cb, pid, err] { cb(pid, err); }));
   } else if (type == kChildExit && len >= sizeof(ExitNotice)) {
     ExitNotice note;
@@ -457,11 +457,11 @@ void Spawner::handleMessage(size_t idx, const char* data, size_t len) {
     }
     if (spawnDebug())
       fprintf(stderr, "[spawner] exit pid %d\n", note.pid);
-    Loop* loop = it->second;
+    auto [loop, loopId] = it->second;
     pidLoops_.erase(it);
     pid_t pid = note.pid;
     int status = note.status;
-    Loop::postIfLive(loop, timedItem("exit", [loop, pid, status] {
+    Loop::postIfLive(loop, loopId, timedItem("exit", [loop, pid, status] {
                        loop->notifyChildExit(pid, status);
                      }));
   }
@@ -495,7 +495,7 @@ void Spawner::helperDied(size_t idx) {
   }
   for (auto& p : lost) {
     SpawnCb cb = std::move(p.cb);
-    Loop::postIfLive(p.loop, [cb] { cb(-1, ECHILD); });
+    Loop::postIfLive(p.loop, p.loopId, [cb] { cb(-1, ECHILD); });
   }
 }
 

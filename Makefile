@@ -24,19 +24,23 @@ clean:
 	rm -rf $(BUILD_DIR) bin containerpilot_amd/_native.so
 
 # Sanitizer builds (evidence in profiles/sanitizers.md)
+# sanitizer binaries stay inside their build dirs (never bin/: that
+# ships to the GPU box and must only hold release builds)
 tsan:
 	cmake -S . -B build-tsan -G Ninja -DCMAKE_BUILD_TYPE=RelWithDebInfo \
+	  -DCPILOT_BIN_IN_TREE=OFF \
 	  -DCMAKE_CXX_FLAGS="-fsanitize=thread -g -O1"
-	ninja -C build-tsan cpilot_unittests containerpilot
-	./bin/cpilot_unittests
-	ninja -C $(BUILD_DIR)  # restore release binaries
+	ninja -C build-tsan cpilot_unittests containerpilot cpilot-spawn-helper
+	CPILOT_SPAWN_HELPER=build-tsan/bin/cpilot-spawn-helper \
+	  ./build-tsan/bin/cpilot_unittests
 
 asan:
 	cmake -S . -B build-asan -G Ninja -DCMAKE_BUILD_TYPE=RelWithDebInfo \
+	  -DCPILOT_BIN_IN_TREE=OFF \
 	  -DCMAKE_CXX_FLAGS="-fsanitize=address,undefined -g -O1"
-	ninja -C build-asan cpilot_unittests containerpilot
-	./bin/cpilot_unittests
-	ninja -C $(BUILD_DIR)
+	ninja -C build-asan cpilot_unittests containerpilot cpilot-spawn-helper
+	CPILOT_SPAWN_HELPER=build-asan/bin/cpilot-spawn-helper \
+	  ./build-asan/bin/cpilot_unittests
 
 soak: build
 	python3 scripts/soak.py 300

@@ -491,25 +491,31 @@ namespace {
 // requests only (TLS sessions and cancellable long-polls stay
 // per-request). At thousands of TTL updates/sec, per-request connects
 // burned measurable CPU on both the daemon and the agent.
-thread_local std::map<std::string, int>* connPool = nullptr;
+struct ConnPool {
+  std::map<std::string, int> conns;
+  // close pooled fds at thread exit: consul workers are joined on
+  // every generation teardown, so a leak here compounds per reload
+  ~ConnPool() {
+    for (auto& kv : conns) close(kv.second);
+  }
+};
+thread_local ConnPool connPool;
 
 int poolTake(const std::string& target) {
-  if (!connPool) return -1;
-  auto it = connPool->find(target);
-  if (it == connPool->end()) return -1;
+  auto it = connPool.conns.find(target);
+  if (it == connPool.conns.end()) return -1;
   int fd = it->second;
-  connPool->erase(it);
+  connPool.conns.erase(it);
   return fd;
 }
 
 void poolStore(const std::string& target, int fd) {
-  if (!connPool) connPool = new std::map<std::string, int>();
-  auto it = connPool->find(target);
-  if (it != connPool->end()) {
+  auto it = connPool.conns.find(target);
+  if (it != connPool.conns.end()) {
     close(it->second);
     it->second = fd;
   } else {
-    (*connPool)[target] = fd;
+    connPool.conns[target] = fd;
   }
 }
 

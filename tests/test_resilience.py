@@ -209,3 +209,41 @@ def test_spawn_helper_death_recovers(daemon_factory):
     assert len(after) == len(before), (before, after)
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_no_fd_leak_across_reloads(daemon_factory, mock_consul):
+    """Reload churn must not leak fds: consul workers are joined per
+    generation and their pooled keep-alive connections must close at
+    thread exit (regression test for the thread_local pool leak)."""
+    import os
+    import time
+
+    mock_consul.set_health("up", [
+        {"ID": "u1", "Address": "10.0.0.1", "Port": 1000}])
+    d = daemon_factory({
+        "consul": mock_consul.address,
+        "stopTimeout": 1,
+        "jobs": [{
+            "name": "app", "exec": "sleep 300", "port": 8000,
+            "interfaces": ["static:127.0.0.1"],
+            "health": {"exec": "true", "interval": 1, "ttl": 5},
+        }],
+        "watches": [{"name": "up", "interval": 1}],
+    }).start()
+    d.wait_for_socket()
+    time.sleep(2)  # let TTL/keep-alive traffic flow
+
+    def fd_count():
+        return len(os.listdir("/proc/%d/fd" % d.proc.pid))
+
+    before = fd_count()
+    for _ in range(8):
+        d.control("POST", "/v3/reload")
+        time.sleep(0.6)
+        d.wait_for_socket(timeout=15)
+        time.sleep(0.6)
+    after = fd_count()
+    assert after <= before + 4, (
+        "fd count grew across reloads: %d -> %d" % (before, after))
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -33,8 +33,8 @@ std::function<void()> timedItem(const char* tag, std::function<void()> fn);
 void dumpItemTags();
 
 // Reset the calling thread to default scheduling (SCHED_OTHER, nice 0).
-// The daemon's reactor thread runs elevated (SCHED_RR or negative nice);
-// background threads and child processes must not inherit that.
+// Defensive: background threads and spawned children must not inherit
+// any elevated scheduling an operator may have started the daemon with.
 void resetThreadScheduling();
 
 class Loop {

@@ -218,24 +218,13 @@ void App::teardownGeneration() {
 }
 
 int App::run() {
-  // Best-effort scheduling boost for the reactor thread: under load the
-  // daemon's own children (thousands of short-lived checks/sec) compete
-  // with the loop for CPU, and scheduling delay lands directly in the
-  // publish->dispatch tail. SCHED_RR keeps the reactor ahead of its own
-  // workload (it sleeps in epoll otherwise); fall back to a niceness
-  // boost, silently, when the caps aren't granted. CPILOT_NO_RT=1 opts
-  // out.
-  if (!cpilotDebugEnv("CPILOT_NO_RT")) {
-    struct sched_param sp;
-    memset(&sp, 0, sizeof(sp));
-    sp.sched_priority = 10;
-    if (sched_setscheduler(0, SCHED_RR, &sp) != 0) {
-      errno = 0;
-      if (nice(-10) == -1 && errno != 0) {
-        // neither granted: run at default priority
-      }
-    }
-  }
+  // NOTE on scheduling: an experiment ran the reactor thread at
+  // SCHED_RR to shield its dispatch tail from the daemon's own child
+  // churn. Measured result: where RT was actually granted it CUT
+  // throughput ~4x under load (the RT loop starves the reader/helper
+  // threads it depends on and trips RT throttling windows), and on the
+  // target boxes CAP_SYS_NICE isn't granted anyway. The daemon runs at
+  // default priority.
   // long-running daemon hygiene: cap glibc arenas (worker threads
   // otherwise each grow their own) and periodically return freed pages
   // to the OS so RSS tracks live data instead of allocator high-water.

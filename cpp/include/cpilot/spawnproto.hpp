@@ -51,8 +51,12 @@ struct ExitNotice {
   int32_t status;  // waitpid status
 };
 
-// requests larger than this are rejected client-side with E2BIG
-constexpr size_t kMaxRequestBytes = 256 * 1024;
+// Requests larger than this are rejected client-side with E2BIG.
+// 1 MiB fits the socketpair SNDBUF as a single SEQPACKET datagram and
+// covers any realistic environment (the kernel's own execve limit is
+// ~2 MiB total; environments that large fail there in the reference
+// too).
+constexpr size_t kMaxRequestBytes = 1024 * 1024;
 
 }  // namespace spawnproto
 }  // namespace cpilot

@@ -247,3 +247,36 @@ def test_no_fd_leak_across_reloads(daemon_factory, mock_consul):
         "fd count grew across reloads: %d -> %d" % (before, after))
     d.terminate()
     assert d.wait(timeout=30) == 0
+
+
+def test_oversized_environment_fails_loudly(daemon_factory):
+    """A spawn whose serialized request exceeds the helper protocol cap
+    (256 KiB) fails as ExitFailed instead of wedging the job (E2BIG
+    path in the spawner client)."""
+    import time
+
+    # ~1.2 MiB of environment split across vars (a single var would
+    # exceed the kernel's per-string exec limit before reaching us);
+    # over the helper protocol's 1 MiB request cap
+    bigenv = {"CPILOT_TEST_BIG_%03d" % i: "x" * 2048 for i in range(600)}
+    d = daemon_factory({
+        "consul": "localhost:79",
+        "stopTimeout": 1,
+        "logging": {"level": "DEBUG"},
+        "jobs": [
+            {"name": "main-app", "exec": ["sleep", "30"]},
+            {"name": "bigenv", "exec": ["true"],
+             "when": {"interval": "500ms"}, "restarts": "unlimited"},
+        ],
+    }, env=bigenv).start()
+    d.wait_for_socket()
+    time.sleep(2)
+    log = d.log()
+    # the spawns failed loudly (E2BIG = "Argument list too long")...
+    assert "unable to start" in log and "{ExitFailed bigenv}" in log, \
+        log[-2000:]
+    # ...and the daemon stayed healthy
+    status, _ = d.control("GET", "/v3/ping")
+    assert status == 200
+    d.terminate()
+    assert d.wait(timeout=30) == 0

@@ -34,10 +34,6 @@
 
 namespace cpilot {
 
-// debug: caller stamps this right before Spawner::spawn; spawn() entry
-// measures the call gap (argument passing) against it
-extern thread_local Clock::time_point g_spawnCallT0;
-
 class Spawner {
  public:
   // callback runs ON THE LOOP THREAD: (pid, errno) — pid < 0 on failure

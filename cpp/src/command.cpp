@@ -208,20 +208,6 @@ void Command::run(Loop& loop, std::shared_ptr<Bus> bus) {
 void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   loop_ = &loop;
   LOG_DEBUG("%s.Run start", name_.c_str());
-  static const bool spdbg = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
-  static double accPipe = 0, accEnv = 0, accEnq = 0;
-  static uint64_t accN = 0;
-  static TimePoint accLast = Clock::now();
-  auto tstart = Clock::now();
-  auto lap = [&](const char* what, double* slot) {
-    if (!spdbg) return;
-    auto now = Clock::now();
-    auto ms = std::chrono::duration<double, std::milli>(now - tstart).count();
-    *slot += ms;
-    if (ms > 5.0)
-      fprintf(stderr, "spawn-step stall: %s after %.1f ms\n", what, ms);
-    tstart = now;
-  };
 
   int pipefds[2] = {-1, -1};
   if (!raw_) {
@@ -234,7 +220,6 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
     }
   }
 
-  lap("pipe2", &accPipe);
   running_ = true;
   pid_ = -1;
   pendingSignal_ = 0;
@@ -244,7 +229,6 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
   // snapshot env (+ the PID-var overlay) on this (the loop) thread; the
   // spawner must never read the live environ concurrently with setenv
   auto baseEnv = baseEnvSnapshot();
-  lap("envSnapshot", &accEnv);
   // the spawner pool does the posix_spawnp so a burst of launches never
   // blocks event dispatch; completion lands back on the loop
   Spawner::SpawnCb compCb =
@@ -320,37 +304,10 @@ void Command::spawn(Loop& loop, std::shared_ptr<Bus> bus) {
           onExit(*loop_, bus, status);
         });
       };
-  lap("lambdaBuild", &accEnv);
-  if (spdbg) g_spawnCallT0 = Clock::now();
   Spawner::global().spawn(loop, exec_, args_, std::move(baseEnv),
                           overlayExtras(), raw_ ? -1 : pipefds[1],
                           std::move(compCb));
-  lap("spawnerEnqueue", &accEnq);
-  if (spdbg) {
-    accN++;
-    if (Clock::now() - accLast > std::chrono::seconds(5)) {
-      fprintf(stderr,
-              "spawn spans: n=%llu pipe2=%.0fms env=%.0fms enqueue=%.0fms\n",
-              (unsigned long long)accN, accPipe, accEnv, accEnq);
-      accPipe = accEnv = accEnq = 0;
-      accN = 0;
-      accLast = Clock::now();
-    }
-  }
 }
-
-namespace {
-// CPILOT_LOOP_DEBUG: accumulated per-span cost inside the exit path
-struct ExitSpans {
-  double logdrain = 0, overlay = 0, publish = 0, respawn = 0;
-  uint64_t n = 0;
-  TimePoint last = Clock::now();
-};
-ExitSpans& exitSpans() {
-  static ExitSpans spans;
-  return spans;
-}
-}  // namespace
 
 void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
   LOG_DEBUG("%s.Run end", name_.c_str());
@@ -368,13 +325,6 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
               name_.c_str(), total, total - execPhase, execPhase);
     }
   }
-  TimePoint tSpan = Clock::now();
-  auto lap = [&](double* slot) {
-    if (!rttDebug) return;
-    auto now = Clock::now();
-    *slot += std::chrono::duration<double, std::milli>(now - tSpan).count();
-    tSpan = now;
-  };
   if (timeoutTimer_) {
     loop.cancelTimer(timeoutTimer_);
     timeoutTimer_ = 0;
@@ -399,9 +349,7 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
     logFd_ = -1;
   }
 
-  lap(&exitSpans().logdrain);
   if (!pidEnvName_.empty()) overlayErase(pidEnvName_);
-  lap(&exitSpans().overlay);
 
   running_ = false;
   pid_ = -1;
@@ -419,23 +367,9 @@ void Command::onExit(Loop& loop, std::shared_ptr<Bus> bus, int status) {
                       name_ + ": exit status " + std::to_string(code)});
   }
 
-  lap(&exitSpans().publish);
   if (pendingRun_) {
     pendingRun_ = false;
     spawn(loop, bus);
-  }
-  lap(&exitSpans().respawn);
-  if (rttDebug) {
-    auto& sp = exitSpans();
-    sp.n++;
-    if (Clock::now() - sp.last > std::chrono::seconds(5)) {
-      fprintf(stderr,
-              "exit spans: n=%llu logdrain=%.0fms overlay=%.0fms "
-              "publish=%.0fms respawn=%.0fms\n",
-              (unsigned long long)sp.n, sp.logdrain, sp.overlay, sp.publish,
-              sp.respawn);
-      sp = ExitSpans{};
-    }
   }
 }
 

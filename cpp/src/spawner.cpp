@@ -59,8 +59,6 @@ bool spawnDebug() {
 
 }  // namespace
 
-thread_local Clock::time_point g_spawnCallT0{};
-
 Spawner& Spawner::global() {
   // intentionally leaked: the reader thread and helper processes live
   // for the process lifetime
@@ -190,22 +188,6 @@ void Spawner::spawn(Loop& loop, const std::string& execPath,
   hdr.envc = (uint32_t)((baseEnv ? baseEnv->size() : 0) + extraEnv.size());
   hdr.wantStdio = stdioFd >= 0 ? 1 : 0;
 
-  static double accSer = 0, accMu = 0, accPost = 0, accGap = 0;
-  static uint64_t accCnt = 0;
-  static TimePoint accT = Clock::now();
-  static const bool dbg3 = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
-  TimePoint tphase = Clock::now();
-  if (dbg3 && g_spawnCallT0.time_since_epoch().count() != 0) {
-    accGap += std::chrono::duration<double, std::milli>(
-                  tphase - g_spawnCallT0).count();
-    g_spawnCallT0 = Clock::time_point{};
-  }
-  auto lap3 = [&](double* slot) {
-    if (!dbg3) return;
-    auto now = Clock::now();
-    *slot += std::chrono::duration<double, std::milli>(now - tphase).count();
-    tphase = now;
-  };
   size_t size = sizeof(hdr) + execPath.size() + 1;
   for (auto& a : args) size += a.size() + 1;
   if (baseEnv)
@@ -230,18 +212,11 @@ void Spawner::spawn(Loop& loop, const std::string& execPath,
     for (auto& e : *baseEnv) append(e);
   for (auto& e : extraEnv) append(e);
 
-  lap3(&accSer);
-  TimePoint tLock = Clock::now();
   size_t idx = 0;
   uint32_t reqId = 0;
   bool queued = false;
   {
     std::lock_guard<std::mutex> l(mu_);
-    if (spawnDebug()) {
-      auto ms = std::chrono::duration<double, std::milli>(Clock::now() -
-                                                          tLock).count();
-      if (ms > 5.0) fprintf(stderr, "spawner mu_ wait %.1f ms\n", ms);
-    }
     reqId = nextReqId_++;
     hdr.reqId = reqId;
     memcpy(buf.data(), &hdr, sizeof(hdr));
@@ -268,31 +243,10 @@ void Spawner::spawn(Loop& loop, const std::string& execPath,
       queued = true;
     }
   }
-  lap3(&accMu);
   if (spawnDebug())
     fprintf(stderr, "[spawner] req %u -> helper %zu\n", reqId, idx);
-  static double accQueueMs = 0, accLockMs = 0, accSendMs = 0;
-  static uint64_t nQueued = 0, nDirect = 0;
-  static TimePoint accLast2 = Clock::now();
-  static const bool dbg2 = cpilotDebugEnv("CPILOT_LOOP_DEBUG");
-  if (dbg2 && Clock::now() - accLast2 > std::chrono::seconds(5)) {
-    fprintf(stderr,
-            "spawn branches: queued=%llu direct=%llu queueMs=%.0f "
-            "lockMs=%.0f sendMs=%.0f\n",
-            (unsigned long long)nQueued, (unsigned long long)nDirect,
-            accQueueMs, accLockMs, accSendMs);
-    nQueued = nDirect = 0;
-    accQueueMs = accLockMs = accSendMs = 0;
-    accLast2 = Clock::now();
-  }
   if (queued) {
-    TimePoint tw = Clock::now();
     wakeReader();
-    if (dbg2) {
-      nQueued++;
-      accQueueMs += std::chrono::duration<double, std::milli>(
-                        Clock::now() - tw).count();
-    }
     return;
   }
   // send outside mu_: only this helper's sendMu serializes the syscall,
@@ -300,16 +254,8 @@ void Spawner::spawn(Loop& loop, const std::string& execPath,
   Helper& h = *helpers_[idx];
   bool sent;
   {
-    TimePoint tl = Clock::now();
     std::lock_guard<std::mutex> sl(h.sendMu);
-    TimePoint ts = Clock::now();
     sent = sendRequest(h, buf, stdioFd);
-    if (dbg2) {
-      nDirect++;
-      accLockMs += std::chrono::duration<double, std::milli>(ts - tl).count();
-      accSendMs += std::chrono::duration<double, std::milli>(
-                       Clock::now() - ts).count();
-    }
   }
   if (!sent) {
     // the fd stays open until actually sent
@@ -319,18 +265,6 @@ void Spawner::spawn(Loop& loop, const std::string& execPath,
     return;
   }
   if (stdioFd >= 0) close(stdioFd);
-  lap3(&accPost);
-  if (dbg3) {
-    accCnt++;
-    if (Clock::now() - accT > std::chrono::seconds(5)) {
-      fprintf(stderr, "spawner inner: n=%llu gap=%.0fms serialize=%.0fms "
-              "mu=%.0fms post=%.0fms\n",
-              (unsigned long long)accCnt, accGap, accSer, accMu, accPost);
-      accSer = accMu = accPost = accGap = 0;
-      accCnt = 0;
-      accT = Clock::now();
-    }
-  }
 }
 
 // Send one request datagram (+fd). Returns false on EAGAIN (caller

@@ -34,8 +34,6 @@ p99 = histogram_p99(s0["buckets"], s1["buckets"])
 log = d.log()[marker:]
 stalls = re.findall(r"loop stall: (\w+) fd=-?\d+ took ([0-9.]+) ms", log)
 tstalls = re.findall(r"timer stall: id=\d+ interval_ms=(\d+) took ([0-9.]+) ms", log)
-sstalls = re.findall(r"spawn-step stall: (\w+) after ([0-9.]+) ms", log)
-muw = re.findall(r"spawner mu_ wait ([0-9.]+) ms", log)
 agg = collections.Counter()
 mx = collections.defaultdict(float)
 tot = collections.defaultdict(float)
@@ -59,9 +57,6 @@ print(json.dumps({
     "slow_worst_ms": worst,
     "timer_stalls": len(tstalls),
     "timer_worst_ms": sorted((float(m) for _, m in tstalls))[-5:],
-    "spawn_step_stalls": collections.Counter(w for w, _ in sstalls),
-    "spawn_step_worst_ms": sorted((float(m) for _, m in sstalls))[-5:],
-    "mu_waits": len(muw),
     "slow_rtts": len(rtts),
     "rtt_samples": sorted(rtts, reverse=True)[:8],
     "rtt_mean_spawn_phase": round(sum(r[1] for r in rtts) /
@@ -72,9 +67,6 @@ print(json.dumps({
     "top_fds": re.findall(r"top fd .*", log)[-4:],
     "posted_items": re.findall(r"posted items: .*", log)[-2:],
     "item_tags": re.findall(r"item tag .*", log)[-3:],
-    "exit_spans": re.findall(r"exit spans: .*", log)[-2:],
-    "spawn_spans": re.findall(r"spawn spans: .*", log)[-2:],
-    "spawn_branches": re.findall(r"spawn branches: .*", log)[-1:],
-    "spawner_inner": re.findall(r"spawner inner: .*", log)[-2:],
+
 }), flush=True)
 d.cleanup()

@@ -6,8 +6,13 @@
 #include <arpa/inet.h>
 #include <netinet/in.h>
 
+#include <poll.h>
+
+#include <atomic>
 #include <cassert>
 #include <cstdio>
+#include <thread>
+#include <vector>
 #include <cstdlib>
 #include <cstring>
 #include <string>
@@ -22,6 +27,7 @@
 #include "cpilot/jobs.hpp"
 #include "cpilot/decode.hpp"
 #include "cpilot/events.hpp"
+#include "cpilot/http.hpp"
 #include "cpilot/ips.hpp"
 #include "cpilot/json.hpp"
 #include "cpilot/timing.hpp"
@@ -729,6 +735,185 @@ static void testStartTimeoutQuitsJob() {
   CHECK_EQ(probe.count(EventCode::Stopped, "patient"), 1);
 }
 
+// ---- HTTP client framing + keep-alive pool (http.cpp) ----
+//
+// A scripted in-process server: accepts connections on a TCP port and
+// answers each request with the next canned raw response, recording how
+// many CONNECTIONS it saw — which is what proves (or disproves) reuse.
+namespace {
+
+struct ScriptedServer {
+  int listenFd = -1;
+  int port = 0;
+  std::vector<std::string> responses;
+  std::atomic<int> connections{0};
+  std::atomic<int> requests{0};
+  std::atomic<bool> stop{false};
+  std::thread thread;
+
+  void start() {
+    listenFd = socket(AF_INET, SOCK_STREAM, 0);
+    int one = 1;
+    setsockopt(listenFd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    struct sockaddr_in addr;
+    memset(&addr, 0, sizeof(addr));
+    addr.sin_family = AF_INET;
+    addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+    CHECK(bind(listenFd, (struct sockaddr*)&addr, sizeof(addr)) == 0);
+    socklen_t len = sizeof(addr);
+    getsockname(listenFd, (struct sockaddr*)&addr, &len);
+    port = ntohs(addr.sin_port);
+    CHECK(listen(listenFd, 8) == 0);
+    thread = std::thread([this] { serve(); });
+  }
+
+  void serve() {
+    while (!stop) {
+      struct pollfd p{listenFd, POLLIN, 0};
+      if (poll(&p, 1, 100) <= 0) continue;
+      int fd = accept(listenFd, nullptr, nullptr);
+      if (fd < 0) continue;
+      struct timeval tv{0, 200000};  // shutdown() must not hang in recv
+      setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+      connections++;
+      // serve requests on this connection until close or script end
+      while (!stop) {
+        char buf[8192];
+        std::string req;
+        bool gotReq = false;
+        bool peerGone = false;
+        while (req.find("\r\n\r\n") == std::string::npos) {
+          ssize_t n = recv(fd, buf, sizeof(buf), 0);
+          if (n == 0) {
+            peerGone = true;
+            break;
+          }
+          if (n < 0) {
+            if ((errno == EAGAIN || errno == EWOULDBLOCK) && !stop) continue;
+            peerGone = true;
+            break;
+          }
+          req.append(buf, n);
+          gotReq = true;
+        }
+        if (peerGone) break;
+        if (!gotReq || req.find("\r\n\r\n") == std::string::npos) break;
+        int idx = requests++;
+        if (idx >= (int)responses.size()) break;
+        const std::string& resp = responses[idx];
+        size_t off = 0;
+        while (off < resp.size()) {
+          ssize_t n = send(fd, resp.data() + off, resp.size() - off,
+                           MSG_NOSIGNAL);
+          if (n <= 0) break;
+          off += n;
+        }
+        if (resp.find("Connection: close") != std::string::npos) break;
+      }
+      close(fd);
+    }
+  }
+
+  void shutdown() {
+    stop = true;
+    if (thread.joinable()) thread.join();
+    if (listenFd >= 0) close(listenFd);
+  }
+};
+
+}  // namespace
+
+static void testHttpClientFraming() {
+  using cpilot::http::request;
+  // content-length framing + keep-alive reuse: three requests over ONE
+  // connection (the pool holds it between calls)
+  {
+    ScriptedServer srv;
+    std::string ok =
+        "HTTP/1.1 200 OK\r\nContent-Type: text/plain\r\n"
+        "Content-Length: 5\r\n\r\nhello";
+    srv.responses = {ok, ok, ok};
+    srv.start();
+    std::string target = "127.0.0.1:" + std::to_string(srv.port);
+    for (int i = 0; i < 3; i++) {
+      auto r = request(target, "GET", "/x", "");
+      CHECK(r.ok);
+      CHECK_EQ(r.status, 200);
+      CHECK_EQ(r.body, std::string("hello"));
+    }
+    CHECK_EQ(srv.connections.load(), 1);  // reused, not re-connected
+    CHECK_EQ(srv.requests.load(), 3);
+    srv.shutdown();
+  }
+  // chunked transfer-encoding is decoded and the connection still reused
+  {
+    ScriptedServer srv;
+    std::string chunked =
+        "HTTP/1.1 200 OK\r\nTransfer-Encoding: chunked\r\n\r\n"
+        "5\r\nhello\r\n6\r\n world\r\n0\r\n\r\n";
+    srv.responses = {chunked, chunked};
+    srv.start();
+    std::string target = "127.0.0.1:" + std::to_string(srv.port);
+    auto r1 = request(target, "GET", "/c", "");
+    CHECK(r1.ok);
+    CHECK_EQ(r1.body, std::string("hello world"));
+    auto r2 = request(target, "GET", "/c", "");
+    CHECK(r2.ok);
+    CHECK_EQ(r2.body, std::string("hello world"));
+    CHECK_EQ(srv.connections.load(), 1);
+    srv.shutdown();
+  }
+  // Connection: close is honored — the next request reconnects
+  {
+    ScriptedServer srv;
+    std::string closing =
+        "HTTP/1.1 200 OK\r\nContent-Length: 2\r\n"
+        "Connection: close\r\n\r\nok";
+    srv.responses = {closing, closing};
+    srv.start();
+    std::string target = "127.0.0.1:" + std::to_string(srv.port);
+    CHECK(request(target, "GET", "/a", "").ok);
+    CHECK(request(target, "GET", "/a", "").ok);
+    CHECK_EQ(srv.connections.load(), 2);
+    srv.shutdown();
+  }
+  // stale pooled connection: server restarts between calls; the client
+  // retries transparently on a fresh connection
+  {
+    std::string ok =
+        "HTTP/1.1 200 OK\r\nContent-Length: 1\r\n\r\nA";
+    int port;
+    {
+      ScriptedServer srv;
+      srv.responses = {ok};
+      srv.start();
+      port = srv.port;
+      std::string target = "127.0.0.1:" + std::to_string(port);
+      CHECK(request(target, "GET", "/s", "").ok);  // pools the conn
+      srv.shutdown();                              // kills it server-side
+    }
+    ScriptedServer srv2;
+    srv2.responses = {ok};
+    srv2.listenFd = socket(AF_INET, SOCK_STREAM, 0);
+    int one = 1;
+    setsockopt(srv2.listenFd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    struct sockaddr_in addr;
+    memset(&addr, 0, sizeof(addr));
+    addr.sin_family = AF_INET;
+    addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+    addr.sin_port = htons(port);
+    CHECK(bind(srv2.listenFd, (struct sockaddr*)&addr, sizeof(addr)) == 0);
+    CHECK(listen(srv2.listenFd, 8) == 0);
+    srv2.port = port;
+    srv2.thread = std::thread([&srv2] { srv2.serve(); });
+    std::string target = "127.0.0.1:" + std::to_string(port);
+    auto r = request(target, "GET", "/s", "");
+    CHECK(r.ok);
+    CHECK_EQ(r.body, std::string("A"));
+    srv2.shutdown();
+  }
+}
+
 int main() {
   testJson5();
   testDurations();
@@ -748,6 +933,7 @@ int main() {
   testMaintenanceMatrix();
   testSignalJobMatrix();
   testStartTimeoutQuitsJob();
+  testHttpClientFraming();
   if (failures) {
     fprintf(stderr, "%d failures\n", failures);
     return 1;

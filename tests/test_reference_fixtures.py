@@ -107,3 +107,52 @@ def test_app_testcfg_shapes():
     err = native.validate_config(
         '{"consul": "consul:8500", "watches": [{"name": "name"}]}')
     assert err and "interval must be > 0" in err
+
+
+
+class TestReferenceDocsExamples:
+    """Run the reference's own docs example configs
+    (docs/30-configuration/examples/*.json5) through our
+    render+validate pipeline and pin the exact outcome of each.
+
+    Notably, most of those examples are broken in the reference's OWN
+    pipeline too (missing JSON5 commas, `health` without `interval` —
+    which jobs/config.go:305 rejects with the same message we emit, a
+    `tll` typo that strict decoding rejects): doc rot the rebuild
+    faithfully rejects the same way rather than silently accepting."""
+
+    EXDIR = os.path.join(REF, "docs", "30-configuration", "examples")
+
+    def _outcome(self, name):
+        from containerpilot_amd import native
+        os.environ.setdefault("CONSUL", "consul.example.com")
+        text = open(os.path.join(self.EXDIR, name)).read()
+        rendered = native.render_template(text)
+        return native.validate_config(rendered)
+
+    def test_examples_present(self):
+        assert sorted(os.listdir(self.EXDIR)) == [
+            "consul-agent.json5", "database-config.json5",
+            "nginx-upstreams.json5", "periodic-tasks.json5",
+            "service-reg-only.json5", "stopping.json5"]
+
+    def test_missing_comma_examples_rejected_at_parse(self):
+        # flynn/json5 (the reference's parser) also requires commas
+        for name in ("consul-agent.json5", "nginx-upstreams.json5",
+                     "periodic-tasks.json5"):
+            err = self._outcome(name)
+            assert err is not None and "parse error" in err, (name, err)
+
+    def test_health_without_interval_rejected_same_message(self):
+        # jobs/config.go:305: job[%s].health.interval must be > 0
+        for name in ("database-config.json5", "service-reg-only.json5"):
+            err = self._outcome(name)
+            assert err is not None and \
+                "job[consul-agent].health.interval must be > 0" in err, \
+                (name, err)
+
+    def test_unknown_key_typo_rejected(self):
+        # stopping.json5 has `tll:` (sic); strict decoding rejects it
+        # like the reference's mapstructure ErrorUnused
+        err = self._outcome("stopping.json5")
+        assert err is not None and "tll" in err, err

@@ -674,8 +674,10 @@ ClientResult request(const std::string& target, const std::string& method,
       std::string decoded;
       while (!tryDechunk(resp.substr(headerEnd + 4), &decoded)) {
         if (eof || !readMore()) {
-          bodyFramed = false;
-          break;
+          // timeout/EOF mid-body: a truncated framed body is an error,
+          // not a short success (Go's client errors the same way)
+          result.error = "truncated response body";
+          return result;
         }
       }
       result.body = decoded;
@@ -684,12 +686,11 @@ ClientResult request(const std::string& target, const std::string& method,
           (size_t)atoll(result.headers["content-length"].c_str());
       while (resp.size() < headerEnd + 4 + want) {
         if (eof || !readMore()) {
-          bodyFramed = false;
-          break;
+          result.error = "truncated response body";
+          return result;
         }
       }
-      result.body = resp.substr(headerEnd + 4,
-                                std::min(want, resp.size() - headerEnd - 4));
+      result.body = resp.substr(headerEnd + 4, want);
     } else {
       // no framing: read to EOF, connection not reusable
       while (readMore()) {

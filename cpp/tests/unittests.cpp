@@ -877,6 +877,20 @@ static void testHttpClientFraming() {
     CHECK_EQ(srv.connections.load(), 2);
     srv.shutdown();
   }
+  // truncated framed body (server closes mid-body) is an error, not a
+  // short success
+  {
+    ScriptedServer srv;
+    srv.responses = {
+        "HTTP/1.1 200 OK\r\nContent-Length: 10\r\n"
+        "Connection: close\r\n\r\nhalf"};
+    srv.start();
+    std::string target = "127.0.0.1:" + std::to_string(srv.port);
+    auto r = request(target, "GET", "/t", "");
+    CHECK(!r.ok);
+    CHECK(r.error.find("truncated") != std::string::npos);
+    srv.shutdown();
+  }
   // stale pooled connection: server restarts between calls; the client
   // retries transparently on a fresh connection
   {

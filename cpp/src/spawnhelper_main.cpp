@@ -181,7 +181,7 @@ int cpilotSpawnHelperMain() {
       }
       reapAndNotify(sock);
     }
-    if (fds[0].revents & (POLLIN | POLLHUP)) {
+    if (fds[0].revents & (POLLIN | POLLHUP | POLLERR)) {
       struct iovec iov{buf.data(), buf.size()};
       struct msghdr msg;
       memset(&msg, 0, sizeof(msg));
